@@ -1,0 +1,54 @@
+"""Per-GPU engine configuration.
+
+The knobs reproduce the serving contract the reference pins on its vLLM fleet
+(SURVEY.md §2.3): block size 16 (`simulations/.../constants.py:13-15`),
+max_num_seqs 256 and chunked-prefill token cap (`constants.py:11,16`),
+preemption-by-recompute at 90% KV (`constants.py:17-18`,
+`continous_batching.py:117-131`), adapter slot budget `--max-loras`
+(`vllm-lora-deployment.yaml:44`), gpu memory utilisation 0.95
+(`tools/dynamic-lora-sidecar/deployment.yaml:40`).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+from ..models.config import ModelConfig
+
+
+@dataclass
+class EngineConfig:
+    model: ModelConfig
+    # Paged KV cache (reference sim: BLOCK_SIZE=16, TOTAL_BLOCKS=2810).
+    block_size: int = 16
+    num_kv_blocks: int = 2810  # overridden by profile_memory on GPU
+    gpu_memory_utilization: float = 0.90
+    # Continuous-batching admission caps (reference sim constants.py:11,16).
+    max_num_seqs: int = 256
+    max_num_batched_tokens: int = 4096
+    # Preempt (recompute) newest sequences above this KV utilisation
+    # (reference continous_batching.py:117-131).
+    preempt_kv_threshold: float = 0.90
+    # Multi-LoRA slots (vLLM --max-loras analogue).
+    max_loras: int = 8
+    max_lora_rank: int = 16
+    # Watermark of blocks kept free when admitting prefills.
+    watermark_blocks: int = 8
+    max_model_len: int = 4096
+    # Engine identity.
+    engine_id: str = "engine-0"
+    gpu_id: int = 0
+    enforce_eager: bool = True  # hipGraph capture of the decode step when False
+
+    def __post_init__(self) -> None:
+        if self.block_size <= 0:
+            raise ValueError("block_size must be positive")
+        if self.max_model_len % self.block_size != 0:
+            # round up to a whole number of blocks
+            self.max_model_len = (
+                (self.max_model_len + self.block_size - 1) // self.block_size
+            ) * self.block_size
+
+    @property
+    def kv_tokens_capacity(self) -> int:
+        return self.num_kv_blocks * self.block_size

@@ -1,0 +1,39 @@
+"""Flat-batch metadata handed from the model runner to the attention ops.
+
+One engine step runs ONE model forward over a flat token batch laid out as
+[all prefill-chunk tokens..., all decode tokens...] with no padding.  The
+attention layer consumes this metadata to (a) scatter new K/V into the paged
+cache and (b) run the prefill and decode attention kernels on their
+partitions.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+
+@dataclass
+class BatchMeta:
+    # Tokens: prefill partition first, then decodes.
+    num_prefill_tokens: int
+    num_decode_tokens: int
+    slot_mapping: torch.Tensor  # [T] int64 — cache slot per token
+
+    # Prefill partition (S_p sequences, chunk sizes vary).
+    prefill_query_start_locs: Optional[torch.Tensor] = None  # [S_p+1] int32
+    prefill_context_lens: Optional[torch.Tensor] = None  # [S_p] int32 (AFTER chunk cached)
+    prefill_block_tables: Optional[torch.Tensor] = None  # [S_p, max_blocks] int32
+
+    # Decode partition (S_d == num_decode_tokens sequences).
+    decode_block_tables: Optional[torch.Tensor] = None  # [S_d, max_blocks] int32
+    decode_context_lens: Optional[torch.Tensor] = None  # [S_d] int32
+
+    # Multi-LoRA (SGMV): per-token adapter slot index, -1 = base model only.
+    lora_slot_per_token: Optional[torch.Tensor] = None  # [T] int32
+
+    @property
+    def num_tokens(self) -> int:
+        return self.num_prefill_tokens + self.num_decode_tokens

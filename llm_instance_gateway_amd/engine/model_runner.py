@@ -1,0 +1,201 @@
+"""Model runner: owns the model, the physical paged KV cache, and one-step
+execution (batch build → forward → sample).
+
+MI355X notes: the KV cache is sized against the 288 GB HBM3E pool at engine
+init (`determine_num_kv_blocks`); batches are flat/unpadded; all per-step
+index tensors are built on host and copied once (a single H2D per tensor per
+step).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from ..models.config import ModelConfig
+from .config import EngineConfig
+from .metadata import BatchMeta
+from .sampling import sample_tokens
+from .scheduler import ScheduledSeq, SchedulerOutput
+from .sequence import Sequence
+
+log = logging.getLogger(__name__)
+
+
+def build_model(cfg: ModelConfig, device, dtype, tp=None):
+    if cfg.is_moe:
+        from ..models.mixtral import MixtralForCausalLM
+
+        return MixtralForCausalLM(cfg, device=device, dtype=dtype, tp=tp)
+    from ..models.llama import LlamaForCausalLM
+
+    return LlamaForCausalLM(cfg, device=device, dtype=dtype, tp=tp)
+
+
+def determine_num_kv_blocks(
+    config: EngineConfig, device: torch.device, tp_size: int = 1
+) -> int:
+    """Size the paged KV pool from free HBM after weights, honoring
+    gpu_memory_utilization (the reference fleet runs vLLM at 0.95 —
+    tools/dynamic-lora-sidecar/deployment.yaml:40)."""
+    mc = config.model
+    bytes_per_block = (
+        mc.num_layers
+        * 2
+        * (mc.num_kv_heads // tp_size)
+        * config.block_size
+        * mc.head_dim
+        * 2  # bf16
+    )
+    if device.type == "cuda":
+        free, total = torch.cuda.mem_get_info(device)
+        budget = int(total * config.gpu_memory_utilization) - (total - free)
+        budget = max(budget, 0)
+        n = budget // bytes_per_block
+        return int(n)
+    return config.num_kv_blocks
+
+
+class ModelRunner:
+    def __init__(
+        self,
+        config: EngineConfig,
+        device: str = "cpu",
+        dtype: torch.dtype = torch.bfloat16,
+        tp=None,
+    ) -> None:
+        self.config = config
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.tp = tp
+        mc = config.model
+        log.info("building model %s on %s", mc.name, self.device)
+        self.model = build_model(mc, self.device, dtype, tp)
+        tp_size = tp.world_size if tp is not None else 1
+        if self.device.type == "cuda":
+            n = determine_num_kv_blocks(config, self.device, tp_size)
+            if n <= 0:
+                raise RuntimeError("no HBM left for KV cache after weights")
+            config.num_kv_blocks = min(
+                n, config.num_kv_blocks if config.num_kv_blocks > 0 else n
+            )
+        kvh = mc.num_kv_heads // tp_size
+        shape = (config.num_kv_blocks, kvh, config.block_size, mc.head_dim)
+        self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = [
+            (
+                torch.zeros(shape, dtype=dtype, device=self.device),
+                torch.zeros(shape, dtype=dtype, device=self.device),
+            )
+            for _ in range(mc.num_layers)
+        ]
+        log.info(
+            "KV cache: %d blocks x %d tokens (%.1f GiB)",
+            config.num_kv_blocks,
+            config.block_size,
+            2 * mc.num_layers * shape[0] * kvh * config.block_size * mc.head_dim * 2 / 2**30,
+        )
+
+    # -- batch construction ---------------------------------------------------
+    def _build_batch(self, out: SchedulerOutput):
+        prefills: List[ScheduledSeq] = []
+        decodes: List[ScheduledSeq] = []
+        for ss in out.scheduled:
+            (decodes if ss.num_tokens == 1 else prefills).append(ss)
+
+        bs = self.config.block_size
+        input_ids: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        sample_seqs: List[Sequence] = []
+        sample_idx: List[int] = []
+
+        qsl = [0]
+        p_ctx: List[int] = []
+        p_bts: List[List[int]] = []
+        for ss in prefills:
+            seq = ss.seq
+            c, n = seq.num_computed_tokens, ss.num_tokens
+            toks = seq.all_token_ids[c : c + n]
+            input_ids.extend(toks)
+            positions.extend(range(c, c + n))
+            for pos in range(c, c + n):
+                slots.append(seq.block_table[pos // bs] * bs + pos % bs)
+            qsl.append(qsl[-1] + n)
+            p_ctx.append(c + n)
+            p_bts.append(seq.block_table)
+            if c + n == seq.total_len:
+                sample_seqs.append(seq)
+                sample_idx.append(qsl[-1] - 1)
+
+        num_prefill_tokens = len(input_ids)
+        d_ctx: List[int] = []
+        d_bts: List[List[int]] = []
+        for ss in decodes:
+            seq = ss.seq
+            pos = seq.num_computed_tokens  # == total_len - 1
+            input_ids.append(seq.all_token_ids[pos])
+            positions.append(pos)
+            slots.append(seq.block_table[pos // bs] * bs + pos % bs)
+            d_ctx.append(pos + 1)
+            d_bts.append(seq.block_table)
+            sample_seqs.append(seq)
+            sample_idx.append(len(input_ids) - 1)
+
+        dev = self.device
+
+        def pad_tables(bts: List[List[int]]) -> Optional[torch.Tensor]:
+            if not bts:
+                return None
+            m = max(len(b) for b in bts)
+            t = torch.zeros(len(bts), m, dtype=torch.int32)
+            for i, b in enumerate(bts):
+                t[i, : len(b)] = torch.tensor(b, dtype=torch.int32)
+            return t.to(dev, non_blocking=True)
+
+        meta = BatchMeta(
+            num_prefill_tokens=num_prefill_tokens,
+            num_decode_tokens=len(decodes),
+            slot_mapping=torch.tensor(slots, dtype=torch.int64).to(dev, non_blocking=True),
+            prefill_query_start_locs=(
+                torch.tensor(qsl, dtype=torch.int32).to(dev, non_blocking=True)
+                if prefills
+                else None
+            ),
+            prefill_context_lens=(
+                torch.tensor(p_ctx, dtype=torch.int32).to(dev, non_blocking=True)
+                if prefills
+                else None
+            ),
+            prefill_block_tables=pad_tables(p_bts),
+            decode_block_tables=pad_tables(d_bts),
+            decode_context_lens=(
+                torch.tensor(d_ctx, dtype=torch.int32).to(dev, non_blocking=True)
+                if decodes
+                else None
+            ),
+        )
+        ids_t = torch.tensor(input_ids, dtype=torch.int64).to(dev, non_blocking=True)
+        pos_t = torch.tensor(positions, dtype=torch.int64).to(dev, non_blocking=True)
+        return ids_t, pos_t, meta, sample_seqs, sample_idx, prefills + decodes
+
+    # -- one step --------------------------------------------------------------
+    @torch.no_grad()
+    def execute(self, out: SchedulerOutput) -> Dict[int, int]:
+        """Run one mixed step; returns {seq_id: sampled_token} and advances
+        num_computed_tokens for every scheduled sequence."""
+        if out.is_empty:
+            return {}
+        ids, pos, meta, sample_seqs, sample_idx, ordered = self._build_batch(out)
+        hidden = self.model(ids, pos, self.kv_caches, meta)
+        sampled: Dict[int, int] = {}
+        if sample_seqs:
+            idx = torch.tensor(sample_idx, dtype=torch.int64, device=self.device)
+            logits = self.model.compute_logits(hidden, idx)
+            tokens = sample_tokens(logits, sample_seqs)
+            for seq, tok in zip(sample_seqs, tokens):
+                sampled[seq.seq_id] = tok
+        for ss in ordered:
+            ss.seq.num_computed_tokens += ss.num_tokens
+        return sampled

@@ -1,0 +1,154 @@
+"""Engine ops: dispatch between the CDNA4 HIP extension and torch references.
+
+Policy:
+  * CUDA (= ROCm/HIP) tensors → the in-tree native extension
+    ``llm_instance_gateway_amd._C`` (hand-written gfx950 kernels).  If the
+    extension is missing on a GPU machine the op RAISES — no silent eager
+    fallback (the driver records which .so files GPU processes load).
+  * CPU tensors → the fp32 torch reference (`torch_ref`), which is also the
+    ground truth the kernels are tested against.
+  * ``LIG_FORCE_TORCH_OPS=1`` forces the torch path on GPU — for A/B numerics
+    and perf comparisons only.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from . import torch_ref
+
+_EXT = None
+_EXT_ERR: Optional[Exception] = None
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from llm_instance_gateway_amd import _C  # built in-tree by setup.py
+
+        _EXT = _C
+    except ImportError as e:
+        _EXT_ERR = e
+    return _EXT
+
+
+def native_available() -> bool:
+    return _load_ext() is not None
+
+
+def _use_native(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    if os.environ.get("LIG_FORCE_TORCH_OPS") == "1":
+        return False
+    ext = _load_ext()
+    if ext is None:
+        raise RuntimeError(
+            "llm_instance_gateway_amd._C native extension is not built but a "
+            "GPU tensor was passed; build it with `python setup.py "
+            f"build_ext --inplace` (import error: {_EXT_ERR})"
+        )
+    return True
+
+
+# -- public op surface -------------------------------------------------------
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if _use_native(x):
+        out = torch.empty_like(x)
+        _EXT.rms_norm(out, x, weight, eps)
+        return out
+    return torch_ref.rms_norm(x, weight, eps)
+
+
+def fused_add_rms_norm(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if _use_native(x):
+        # In-place: residual += x; x = rmsnorm(residual) written to out.
+        out = torch.empty_like(x)
+        _EXT.fused_add_rms_norm(out, x, residual, weight, eps)
+        return out, residual
+    return torch_ref.fused_add_rms_norm(x, residual, weight, eps)
+
+
+def rotary_embedding(
+    positions: torch.Tensor,
+    q: torch.Tensor,
+    k: torch.Tensor,
+    cos_sin_cache: torch.Tensor,
+    head_dim: int,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if _use_native(q):
+        _EXT.rotary_embedding(positions, q, k, cos_sin_cache, head_dim)
+        return q, k
+    return torch_ref.rotary_embedding(positions, q, k, cos_sin_cache, head_dim)
+
+
+def reshape_and_cache(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    if _use_native(k):
+        _EXT.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        return
+    torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+def paged_attention_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    if _use_native(q):
+        out = torch.empty_like(q)
+        _EXT.paged_attention_decode(
+            out, q, k_cache, v_cache, block_tables, context_lens, scale
+        )
+        return out
+    return torch_ref.paged_attention_decode(
+        q, k_cache, v_cache, block_tables, context_lens, scale
+    )
+
+
+def paged_attention_prefill(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    query_start_locs: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    if _use_native(q):
+        out = torch.empty_like(q)
+        _EXT.paged_attention_prefill(
+            out, q, k_cache, v_cache, block_tables, query_start_locs, context_lens, scale
+        )
+        return out
+    return torch_ref.paged_attention_prefill(
+        q, k_cache, v_cache, block_tables, query_start_locs, context_lens, scale
+    )
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if _use_native(x):
+        d = x.shape[-1] // 2
+        out = torch.empty(*x.shape[:-1], d, dtype=x.dtype, device=x.device)
+        _EXT.silu_and_mul(out, x)
+        return out
+    return torch_ref.silu_and_mul(x)
+
+
+build_cos_sin_cache = torch_ref.build_cos_sin_cache
