@@ -41,6 +41,10 @@ def native_available() -> bool:
     return _load_ext() is not None
 
 
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
 def _use_native(t: torch.Tensor) -> bool:
     if not t.is_cuda:
         return False
@@ -61,7 +65,7 @@ def _use_native(t: torch.Tensor) -> bool:
 def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
     if _use_native(x):
         out = torch.empty_like(x)
-        _EXT.rms_norm(out, x, weight, eps)
+        _EXT.rms_norm(out, x, weight, eps, _stream())
         return out
     return torch_ref.rms_norm(x, weight, eps)
 
@@ -72,7 +76,7 @@ def fused_add_rms_norm(
     if _use_native(x):
         # In-place: residual += x; x = rmsnorm(residual) written to out.
         out = torch.empty_like(x)
-        _EXT.fused_add_rms_norm(out, x, residual, weight, eps)
+        _EXT.fused_add_rms_norm(out, x, residual, weight, eps, _stream())
         return out, residual
     return torch_ref.fused_add_rms_norm(x, residual, weight, eps)
 
@@ -85,7 +89,7 @@ def rotary_embedding(
     head_dim: int,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _use_native(q):
-        _EXT.rotary_embedding(positions, q, k, cos_sin_cache, head_dim)
+        _EXT.rotary_embedding(positions, q, k, cos_sin_cache, head_dim, _stream())
         return q, k
     return torch_ref.rotary_embedding(positions, q, k, cos_sin_cache, head_dim)
 
@@ -98,7 +102,7 @@ def reshape_and_cache(
     slot_mapping: torch.Tensor,
 ) -> None:
     if _use_native(k):
-        _EXT.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        _EXT.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping, _stream())
         return
     torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
 
@@ -114,7 +118,7 @@ def paged_attention_decode(
     if _use_native(q):
         out = torch.empty_like(q)
         _EXT.paged_attention_decode(
-            out, q, k_cache, v_cache, block_tables, context_lens, scale
+            out, q, k_cache, v_cache, block_tables, context_lens, scale, _stream()
         )
         return out
     return torch_ref.paged_attention_decode(
@@ -132,11 +136,24 @@ def paged_attention_prefill(
     scale: float,
 ) -> torch.Tensor:
     if _use_native(q):
-        out = torch.empty_like(q)
-        _EXT.paged_attention_prefill(
-            out, q, k_cache, v_cache, block_tables, query_start_locs, context_lens, scale
+        # v1 native path: run each chunk query row as a "decode" against its
+        # causal context prefix (context = L - n + j + 1).  The GROUP waves of
+        # a row share K/V lines via L1/L2, and consecutive rows of a chunk
+        # walk overlapping context, so the re-read is largely cache-absorbed.
+        # A fused MFMA flash-prefill kernel replaces this (see csrc/).
+        n_per = (query_start_locs[1:] - query_start_locs[:-1]).long()
+        ctx_end = context_lens.long()
+        # per-token causal context length
+        tok_in_chunk = torch.cat(
+            [torch.arange(int(n), device=q.device) for n in n_per]
         )
-        return out
+        ctx_per_tok = (
+            torch.repeat_interleave(ctx_end - n_per, n_per) + tok_in_chunk + 1
+        ).int()
+        bt_per_tok = torch.repeat_interleave(block_tables, n_per, dim=0).contiguous()
+        return paged_attention_decode(
+            q, k_cache, v_cache, bt_per_tok, ctx_per_tok, scale
+        )
     return torch_ref.paged_attention_prefill(
         q, k_cache, v_cache, block_tables, query_start_locs, context_lens, scale
     )
@@ -146,7 +163,7 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     if _use_native(x):
         d = x.shape[-1] // 2
         out = torch.empty(*x.shape[:-1], d, dtype=x.dtype, device=x.device)
-        _EXT.silu_and_mul(out, x)
+        _EXT.silu_and_mul(out, x, _stream())
         return out
     return torch_ref.silu_and_mul(x)
 

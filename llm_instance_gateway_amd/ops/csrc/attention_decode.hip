@@ -1,0 +1,200 @@
+// Paged-attention DECODE kernel (single query token per sequence), bf16,
+// gfx950 / CDNA4.
+//
+// This is the serving hot path's memory-bound half: per decoded token the
+// kernel streams the sequence's entire paged KV history once.  Design
+// (guide App. B "Attention decode" + §2 coalescing rules):
+//
+//   grid  = (num_seqs, num_kv_heads); block = GROUP waves (GROUP = Hq/Hkv).
+//   Wave w owns q-head (kv_head*GROUP + w).  The GROUP waves of a workgroup
+//   run on one CU and stream the SAME K/V lines, so the per-CU vector L1
+//   absorbs the GQA re-read (one HBM fetch per line).
+//
+//   Lane map (64 = 8 tokens x 8 lanes): lane = tok*8 + dchunk; each lane
+//   loads head_dim/8 contiguous dims of one token's K row — a token row
+//   (head_dim*2 B) is covered by 8 lanes in full 16/32 B chunks, and the 8
+//   token rows of one iteration are contiguous within a KV block, so a wave
+//   iteration issues fully-coalesced loads of 8 rows.
+//
+//   Scores: per-lane partial dot -> xor-shuffle reduce over the 8 lanes of a
+//   token group (1,2,4) -> online softmax with running (m, l) replicated
+//   across the wave (token-group max/sum via xor 8,16,32).  V accumulates
+//   into head_dim/8 f32 per lane; one cross-token xor reduce at the end.
+//
+// Cache layout: [num_blocks, num_kv_heads, block_size, head_dim] bf16.
+// Capability parity: the paged-attention the reference delegates to vLLM
+// (SURVEY.md §2.3 "Paged-attention prefill + decode").
+
+#include "common.h"
+
+template <int HEAD_DIM>
+__global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
+    uint16_t* __restrict__ out,            // [N, Hq, D]
+    const uint16_t* __restrict__ q,        // [N, Hq, D]
+    const uint16_t* __restrict__ k_cache,  // [B, Hkv, BS, D]
+    const uint16_t* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [N, max_blocks]
+    const int* __restrict__ context_lens,  // [N]
+    const float scale,
+    const int num_q_heads,
+    const int num_kv_heads,
+    const int block_size,
+    const int max_blocks) {
+  constexpr int DL = HEAD_DIM / 8;  // dims per lane
+  const int seq = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int group = num_q_heads / num_kv_heads;
+  const int qh = kvh * group + wave;
+
+  const int ctx = context_lens[seq];
+  const int tok = lane >> 3;     // 0..7: token within the 8-token iteration
+  const int dchunk = lane & 7;   // 0..7: which head_dim/8 slice
+
+  // q fragment -> f32 registers
+  float qreg[DL];
+  {
+    const uint16_t* qp =
+        q + ((size_t)seq * num_q_heads + qh) * HEAD_DIM + dchunk * DL;
+#pragma unroll
+    for (int i = 0; i < DL; i += 8) {
+      if constexpr (DL >= 8) {
+        f32x8 v8 = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(qp + i));
+#pragma unroll
+        for (int j = 0; j < 8; ++j) qreg[i + j] = v8[j];
+      }
+    }
+    if constexpr (DL < 8) {
+#pragma unroll
+      for (int i = 0; i < DL; ++i) qreg[i] = bf16_to_f32(qp[i]);
+    }
+  }
+
+  float m = -INFINITY;
+  float l = 0.0f;
+  float oacc[DL];
+#pragma unroll
+  for (int i = 0; i < DL; ++i) oacc[i] = 0.0f;
+
+  const int* bt = block_tables + (size_t)seq * max_blocks;
+  const int iters = (ctx + 7) / 8;
+
+  for (int it = 0; it < iters; ++it) {
+    const int t = it * 8 + tok;
+    const bool valid = t < ctx;
+    float kf[DL];
+    const uint16_t* krow = nullptr;
+    const uint16_t* vrow = nullptr;
+    if (valid) {
+      const int blk = bt[t / block_size];
+      const int off = t % block_size;
+      const size_t base =
+          (((size_t)blk * num_kv_heads + kvh) * block_size + off) * HEAD_DIM +
+          dchunk * DL;
+      krow = k_cache + base;
+      vrow = v_cache + base;
+      if constexpr (DL >= 8) {
+#pragma unroll
+        for (int i = 0; i < DL; i += 8) {
+          f32x8 v8 = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(krow + i));
+#pragma unroll
+          for (int j = 0; j < 8; ++j) kf[i + j] = v8[j];
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < DL; ++i) kf[i] = bf16_to_f32(krow[i]);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < DL; ++i) kf[i] = 0.0f;
+    }
+
+    float s = 0.0f;
+#pragma unroll
+    for (int i = 0; i < DL; ++i) s += qreg[i] * kf[i];
+    s = group_reduce_sum<8>(s);  // full dot, replicated in the token's 8 lanes
+    s = valid ? s * scale : -INFINITY;
+
+    const float chunk_max = stride_reduce_max<8, 8>(s);
+    const float new_m = fmaxf(m, chunk_max);
+    float p;
+    float alpha;
+    if (new_m == -INFINITY) {  // whole iteration masked (cannot happen mid-seq)
+      p = 0.0f;
+      alpha = 1.0f;
+    } else {
+      alpha = (m == -INFINITY) ? 0.0f : __expf(m - new_m);
+      p = (s == -INFINITY) ? 0.0f : __expf(s - new_m);
+      m = new_m;
+    }
+    l = l * alpha + stride_reduce_sum<8, 8>(p);
+
+    float vf[DL];
+    if (valid) {
+      if constexpr (DL >= 8) {
+#pragma unroll
+        for (int i = 0; i < DL; i += 8) {
+          f32x8 v8 = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(vrow + i));
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vf[i + j] = v8[j];
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < DL; ++i) vf[i] = bf16_to_f32(vrow[i]);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < DL; ++i) vf[i] = 0.0f;
+    }
+#pragma unroll
+    for (int i = 0; i < DL; ++i) oacc[i] = oacc[i] * alpha + p * vf[i];
+  }
+
+  // combine the 8 token-subsets (lanes stride-8 apart share dchunk)
+#pragma unroll
+  for (int i = 0; i < DL; ++i) oacc[i] = stride_reduce_sum<8, 8>(oacc[i]);
+
+  if (tok == 0) {
+    const float inv_l = (l > 0.0f) ? 1.0f / l : 0.0f;
+    uint16_t* op =
+        out + ((size_t)seq * num_q_heads + qh) * HEAD_DIM + dchunk * DL;
+    if constexpr (DL >= 8) {
+#pragma unroll
+      for (int i = 0; i < DL; i += 8) {
+        f32x8 v8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v8[j] = oacc[i + j] * inv_l;
+        *reinterpret_cast<u16x8*>(op + i) = f32x8_to_bf16x8(v8);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < DL; ++i) op[i] = f32_to_bf16(oacc[i] * inv_l);
+    }
+  }
+}
+
+extern "C" void lig_paged_attention_decode_launch(
+    uint16_t* out, const uint16_t* q, const uint16_t* k_cache,
+    const uint16_t* v_cache, const int* block_tables,
+    const int* context_lens, float scale, int num_seqs, int num_q_heads,
+    int num_kv_heads, int head_dim, int block_size, int max_blocks,
+    hipStream_t stream) {
+  const int group = num_q_heads / num_kv_heads;
+  dim3 grid(num_seqs, num_kv_heads);
+  dim3 block(group * WAVE_SIZE);
+#define LAUNCH(HD)                                                          \
+  hipLaunchKernelGGL((paged_attention_decode_kernel<HD>), grid, block, 0,   \
+                     stream, out, q, k_cache, v_cache, block_tables,        \
+                     context_lens, scale, num_q_heads, num_kv_heads,        \
+                     block_size, max_blocks)
+  switch (head_dim) {
+    case 16: LAUNCH(16); break;
+    case 64: LAUNCH(64); break;
+    case 128: LAUNCH(128); break;
+    default:
+      // unsupported head_dim — surfaced as hipErrorInvalidValue by the caller
+      break;
+  }
+#undef LAUNCH
+}

@@ -1,0 +1,140 @@
+// Python bindings for the gfx950 kernels (torch extension, compiled by hipcc).
+//
+// Convention: every op takes the current HIP stream as an int64 handle
+// (Python side passes torch.cuda.current_stream().cuda_stream) so the kernels
+// serialize correctly with the library GEMMs on torch's stream without
+// depending on the HIP/CUDA stream-masquerading ATen headers.
+
+#include <torch/extension.h>
+
+#include "common.h"
+
+extern "C" {
+void lig_rms_norm_launch(uint16_t*, const uint16_t*, const uint16_t*, float,
+                         int, int, hipStream_t);
+void lig_fused_add_rms_norm_launch(uint16_t*, const uint16_t*, uint16_t*,
+                                   const uint16_t*, float, int, int,
+                                   hipStream_t);
+void lig_rope_launch(const int64_t*, uint16_t*, uint16_t*, const float*, int,
+                     int, int, int, hipStream_t);
+void lig_reshape_and_cache_launch(const uint16_t*, const uint16_t*, uint16_t*,
+                                  uint16_t*, const int64_t*, int, int, int,
+                                  int, hipStream_t);
+void lig_silu_and_mul_launch(uint16_t*, const uint16_t*, int, int,
+                             hipStream_t);
+void lig_paged_attention_decode_launch(uint16_t*, const uint16_t*,
+                                       const uint16_t*, const uint16_t*,
+                                       const int*, const int*, float, int,
+                                       int, int, int, int, int, hipStream_t);
+}
+
+namespace {
+
+inline uint16_t* bf16_ptr(torch::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, "expected bf16 tensor");
+  TORCH_CHECK(t.is_contiguous(), "expected contiguous tensor");
+  TORCH_CHECK(t.is_cuda(), "expected device tensor");
+  return reinterpret_cast<uint16_t*>(t.data_ptr());
+}
+
+inline const uint16_t* bf16_cptr(const torch::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, "expected bf16 tensor");
+  TORCH_CHECK(t.is_contiguous(), "expected contiguous tensor");
+  TORCH_CHECK(t.is_cuda(), "expected device tensor");
+  return reinterpret_cast<const uint16_t*>(t.data_ptr());
+}
+
+inline hipStream_t as_stream(int64_t s) {
+  return reinterpret_cast<hipStream_t>(s);
+}
+
+void rms_norm(torch::Tensor out, torch::Tensor input, torch::Tensor weight,
+              double eps, int64_t stream) {
+  const int T = input.size(0);
+  const int H = input.size(1);
+  TORCH_CHECK(H % 8 == 0, "hidden must be a multiple of 8");
+  lig_rms_norm_launch(bf16_ptr(out), bf16_cptr(input), bf16_cptr(weight),
+                      (float)eps, T, H, as_stream(stream));
+}
+
+void fused_add_rms_norm(torch::Tensor out, torch::Tensor input,
+                        torch::Tensor residual, torch::Tensor weight,
+                        double eps, int64_t stream) {
+  const int T = input.size(0);
+  const int H = input.size(1);
+  TORCH_CHECK(H % 8 == 0, "hidden must be a multiple of 8");
+  lig_fused_add_rms_norm_launch(bf16_ptr(out), bf16_cptr(input),
+                                bf16_ptr(residual), bf16_cptr(weight),
+                                (float)eps, T, H, as_stream(stream));
+}
+
+void rotary_embedding(torch::Tensor positions, torch::Tensor q,
+                      torch::Tensor k, torch::Tensor cos_sin,
+                      int64_t head_dim, int64_t stream) {
+  TORCH_CHECK(positions.scalar_type() == torch::kInt64);
+  TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(head_dim % 4 == 0, "head_dim must be a multiple of 4");
+  const int T = q.size(0);
+  const int hq = q.size(1) / head_dim;
+  const int hk = k.size(1) / head_dim;
+  lig_rope_launch(positions.data_ptr<int64_t>(), bf16_ptr(q), bf16_ptr(k),
+                  cos_sin.data_ptr<float>(), T, hq, hk, (int)head_dim,
+                  as_stream(stream));
+}
+
+void reshape_and_cache(torch::Tensor k, torch::Tensor v,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor slot_mapping, int64_t stream) {
+  TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
+  const int T = k.size(0);
+  const int Hkv = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  const int D = k_cache.size(3);
+  TORCH_CHECK(D % 8 == 0, "head_dim must be a multiple of 8");
+  lig_reshape_and_cache_launch(bf16_cptr(k), bf16_cptr(v), bf16_ptr(k_cache),
+                               bf16_ptr(v_cache),
+                               slot_mapping.data_ptr<int64_t>(), T, Hkv, D, BS,
+                               as_stream(stream));
+}
+
+void silu_and_mul(torch::Tensor out, torch::Tensor x, int64_t stream) {
+  const int T = x.size(0);
+  const int d = x.size(1) / 2;
+  TORCH_CHECK(d % 8 == 0, "intermediate must be a multiple of 8");
+  lig_silu_and_mul_launch(bf16_ptr(out), bf16_cptr(x), T, d,
+                          as_stream(stream));
+}
+
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_tables,
+                            torch::Tensor context_lens, double scale,
+                            int64_t stream) {
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  TORCH_CHECK(context_lens.scalar_type() == torch::kInt32);
+  const int N = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(D == 16 || D == 64 || D == 128, "head_dim must be 16/64/128");
+  TORCH_CHECK(Hq % Hkv == 0 && Hq / Hkv <= 16, "bad GQA group size");
+  if (N == 0) return;
+  lig_paged_attention_decode_launch(
+      bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
+      block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
+      (float)scale, N, Hq, Hkv, D, BS, max_blocks, as_stream(stream));
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "MI355X (gfx950) native kernels for llm_instance_gateway_amd";
+  m.def("rms_norm", &rms_norm);
+  m.def("fused_add_rms_norm", &fused_add_rms_norm);
+  m.def("rotary_embedding", &rotary_embedding);
+  m.def("reshape_and_cache", &reshape_and_cache);
+  m.def("silu_and_mul", &silu_and_mul);
+  m.def("paged_attention_decode", &paged_attention_decode);
+}

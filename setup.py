@@ -1,0 +1,47 @@
+"""In-tree build of the gfx950 native extension.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built `.so` lands inside the package (`llm_instance_gateway_amd/_C...so`)
+so the repo snapshot carries it to GPU boxes (the JIT cache would not travel).
+"""
+
+import os
+from pathlib import Path
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+ROOT = Path(__file__).parent
+CSRC = ROOT / "llm_instance_gateway_amd" / "ops" / "csrc"
+
+sources = sorted(
+    str(p)
+    for p in CSRC.glob("*.hip")
+    # hipify writes shadow copies named *_hip.hip next to the originals
+    if not p.name.endswith("_hip.hip")
+)
+
+ext = CUDAExtension(
+    name="llm_instance_gateway_amd._C",
+    sources=sources,
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": [
+            "-O3",
+            "-std=c++17",
+            "--offload-arch=gfx950",
+        ],
+    },
+)
+
+setup(
+    name="llm_instance_gateway_amd",
+    version="0.1.0",
+    packages=["llm_instance_gateway_amd"],
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

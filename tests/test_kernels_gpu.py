@@ -1,0 +1,166 @@
+"""GPU numerics tests: every CDNA4 kernel vs the fp32 torch reference
+(kernel unit tests vs PyTorch-ROCm reference outputs)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from llm_instance_gateway_amd import ops
+from llm_instance_gateway_amd.ops import torch_ref
+
+
+DEV = "cuda:0"
+
+
+def setup_module(module):
+    if not ops.native_available():
+        raise RuntimeError(
+            "native extension _C not built — GPU tests must not fall back"
+        )
+
+
+def close(a, b, atol=2e-2, rtol=2e-2):
+    torch.testing.assert_close(a.float(), b.float(), atol=atol, rtol=rtol)
+
+
+@pytest.mark.parametrize("T,H", [(1, 4096), (37, 4096), (256, 1024), (8, 64)])
+def test_rms_norm(T, H):
+    torch.manual_seed(0)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    got = ops.rms_norm(x, w, 1e-5)
+    ref = torch_ref.rms_norm(x, w, 1e-5)
+    close(got, ref)
+
+
+@pytest.mark.parametrize("T,H", [(64, 4096), (3, 128)])
+def test_fused_add_rms_norm(T, H):
+    torch.manual_seed(1)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    r = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    x2, r2 = x.clone(), r.clone()
+    got_y, got_res = ops.fused_add_rms_norm(x, r, w, 1e-5)
+    ref_y, ref_res = torch_ref.fused_add_rms_norm(x2, r2, w, 1e-5)
+    close(got_y, ref_y)
+    close(got_res, ref_res)
+
+
+@pytest.mark.parametrize("T,HQ,HK,D", [(17, 32, 8, 128), (5, 4, 2, 64)])
+def test_rope(T, HQ, HK, D):
+    torch.manual_seed(2)
+    pos = torch.randint(0, 2048, (T,), dtype=torch.int64, device=DEV)
+    q = torch.randn(T, HQ * D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, HK * D, dtype=torch.bfloat16, device=DEV)
+    cache = ops.build_cos_sin_cache(D, 4096, 500000.0, DEV)
+    q2, k2 = q.clone(), k.clone()
+    gq, gk = ops.rotary_embedding(pos, q, k, cache, D)
+    rq, rk = torch_ref.rotary_embedding(pos, q2, k2, cache, D)
+    close(gq, rq)
+    close(gk, rk)
+
+
+def test_reshape_and_cache():
+    torch.manual_seed(3)
+    T, KVH, D, BS, NB = 33, 8, 128, 16, 64
+    k = torch.randn(T, KVH * D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, KVH * D, dtype=torch.bfloat16, device=DEV)
+    kc = torch.zeros(NB, KVH, BS, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    kc2, vc2 = kc.clone(), vc.clone()
+    slots = torch.randperm(NB * BS, device=DEV)[:T].to(torch.int64)
+    ops.reshape_and_cache(k, v, kc, vc, slots)
+    torch_ref.reshape_and_cache(k, v, kc2, vc2, slots)
+    close(kc, kc2, atol=0, rtol=0)
+    close(vc, vc2, atol=0, rtol=0)
+
+
+@pytest.mark.parametrize("T,I", [(64, 14336), (7, 128)])
+def test_silu_and_mul(T, I):
+    torch.manual_seed(4)
+    x = torch.randn(T, 2 * I, dtype=torch.bfloat16, device=DEV)
+    close(ops.silu_and_mul(x), torch_ref.silu_and_mul(x))
+
+
+def _rand_cache_and_tables(NS, KVH, D, BS, max_ctx, seed=5):
+    torch.manual_seed(seed)
+    max_blocks = (max_ctx + BS - 1) // BS
+    NB = NS * max_blocks + 1
+    kc = torch.randn(NB, KVH, BS, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(NB, KVH, BS, D, dtype=torch.bfloat16, device=DEV)
+    perm = torch.randperm(NB - 1)[: NS * max_blocks] + 1
+    bt = perm.view(NS, max_blocks).to(torch.int32).to(DEV)
+    ctx = torch.randint(1, max_ctx + 1, (NS,), dtype=torch.int32, device=DEV)
+    ctx[0] = max_ctx  # pin one long one
+    return kc, vc, bt, ctx
+
+
+@pytest.mark.parametrize(
+    "NS,HQ,KVH,D,BS,max_ctx",
+    [
+        (4, 32, 8, 128, 16, 300),
+        (16, 32, 8, 128, 16, 1024),
+        (3, 4, 2, 64, 16, 77),
+        (2, 4, 2, 16, 16, 33),
+        (1, 8, 8, 128, 16, 1),  # MHA, single token
+    ],
+)
+def test_paged_attention_decode(NS, HQ, KVH, D, BS, max_ctx):
+    kc, vc, bt, ctx = _rand_cache_and_tables(NS, KVH, D, BS, max_ctx)
+    q = torch.randn(NS, HQ, D, dtype=torch.bfloat16, device=DEV)
+    scale = D ** -0.5
+    got = ops.paged_attention_decode(q, kc, vc, bt, ctx, scale)
+    ref = torch_ref.paged_attention_decode(q, kc, vc, bt, ctx, scale)
+    close(got, ref)
+
+
+def test_paged_attention_prefill_matches_ref():
+    NS, HQ, KVH, D, BS = 3, 32, 8, 128, 16
+    max_ctx = 256
+    kc, vc, bt, ctx = _rand_cache_and_tables(NS, KVH, D, BS, max_ctx, seed=7)
+    # chunk sizes n_i <= ctx_i
+    ns = [min(int(ctx[i]), 40 + 13 * i) for i in range(NS)]
+    qsl = torch.tensor([0] + list(torch.tensor(ns).cumsum(0)), dtype=torch.int32, device=DEV)
+    q = torch.randn(int(qsl[-1]), HQ, D, dtype=torch.bfloat16, device=DEV)
+    scale = D ** -0.5
+    got = ops.paged_attention_prefill(q, kc, vc, bt, qsl, ctx, scale)
+    ref = torch_ref.paged_attention_prefill(q, kc, vc, bt, qsl, ctx, scale)
+    close(got, ref)
+
+
+def test_engine_e2e_gpu_matches_cpu_greedy():
+    """Tiny-model generation on GPU (all custom kernels) must match the CPU
+    torch-reference path token-for-token under greedy sampling."""
+    from llm_instance_gateway_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from llm_instance_gateway_amd.models.config import get_model_config
+
+    torch.manual_seed(0)
+    cfg = dict(
+        model=get_model_config("tiny"),
+        num_kv_blocks=128,
+        max_num_seqs=8,
+        max_num_batched_tokens=64,
+        max_model_len=256,
+    )
+    cpu = LLMEngine(EngineConfig(**cfg), device="cpu")
+    gpu = LLMEngine(EngineConfig(**cfg), device=DEV)
+    gpu.runner.model.load_state_dict(
+        {k: v.to(DEV) for k, v in cpu.runner.model.state_dict().items()}
+    )
+    g = torch.Generator().manual_seed(11)
+    prompts = [torch.randint(1, 500, (24,), generator=g).tolist() for _ in range(3)]
+    sp = SamplingParams(max_new_tokens=8)
+    cseqs = [cpu.add_request(p, sp) for p in prompts]
+    gseqs = [gpu.add_request(p, sp) for p in prompts]
+    cpu.generate_all()
+    gpu.generate_all()
+    # bf16 kernel vs fp32-ref divergence can flip an argmax on a random-init
+    # model; require the large majority of tokens to agree.
+    agree = sum(
+        int(a == b)
+        for cs, gs in zip(cseqs, gseqs)
+        for a, b in zip(cs.output_token_ids, gs.output_token_ids)
+    )
+    total = sum(len(cs.output_token_ids) for cs in cseqs)
+    assert agree >= total * 0.7, f"only {agree}/{total} tokens agree"
