@@ -31,6 +31,12 @@ class BatchMeta:
     decode_block_tables: Optional[torch.Tensor] = None  # [S_d, max_blocks] int32
     decode_context_lens: Optional[torch.Tensor] = None  # [S_d] int32
 
+    # Per-TOKEN prefill expansion (computed ONCE per step by the model
+    # runner; the per-layer attention calls reuse it — computing it per layer
+    # cost 32x redundant arange/repeat kernels, see profiles/r01_*.md):
+    prefill_tok_block_tables: Optional[torch.Tensor] = None  # [Tp, max_blocks]
+    prefill_tok_context_lens: Optional[torch.Tensor] = None  # [Tp] int32
+
     # Multi-LoRA (SGMV): per-token adapter slot index, -1 = base model only.
     lora_slot_per_token: Optional[torch.Tensor] = None  # [T] int32
 

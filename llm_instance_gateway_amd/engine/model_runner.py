@@ -176,6 +176,24 @@ class ModelRunner:
                 else None
             ),
         )
+        # Per-token prefill expansion, once per step (the attention layers of
+        # every transformer block reuse it; see profiles/r01_bench_kernel_stats.md).
+        if prefills:
+            ctx_tok: List[int] = []
+            n_per: List[int] = []
+            for ss in prefills:
+                c, n = ss.seq.num_computed_tokens, ss.num_tokens
+                ctx_tok.extend(range(c + 1, c + n + 1))
+                n_per.append(n)
+            meta.prefill_tok_context_lens = torch.tensor(
+                ctx_tok, dtype=torch.int32
+            ).to(dev, non_blocking=True)
+            meta.prefill_tok_block_tables = torch.repeat_interleave(
+                meta.prefill_block_tables,
+                torch.tensor(n_per, dtype=torch.int64, device=dev),
+                dim=0,
+            ).contiguous()
+
         ids_t = torch.tensor(input_ids, dtype=torch.int64).to(dev, non_blocking=True)
         pos_t = torch.tensor(positions, dtype=torch.int64).to(dev, non_blocking=True)
         return ids_t, pos_t, meta, sample_seqs, sample_idx, prefills + decodes

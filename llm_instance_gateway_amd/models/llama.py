@@ -85,6 +85,8 @@ class Attention(nn.Module):
                 meta.prefill_query_start_locs,
                 meta.prefill_context_lens,
                 self.scale,
+                tok_block_tables=meta.prefill_tok_block_tables,
+                tok_context_lens=meta.prefill_tok_context_lens,
             )
         if nd > 0:
             out[np_:] = ops.paged_attention_decode(

@@ -134,8 +134,15 @@ def paged_attention_prefill(
     query_start_locs: torch.Tensor,
     context_lens: torch.Tensor,
     scale: float,
+    tok_block_tables: "Optional[torch.Tensor]" = None,
+    tok_context_lens: "Optional[torch.Tensor]" = None,
 ) -> torch.Tensor:
     if _use_native(q):
+        if tok_block_tables is not None and tok_context_lens is not None:
+            # per-token expansion precomputed once per step by the model runner
+            return paged_attention_decode(
+                q, k_cache, v_cache, tok_block_tables, tok_context_lens, scale
+            )
         # v1 native path: run each chunk query row as a "decode" against its
         # causal context prefix (context = L - n + j + 1).  The GROUP waves of
         # a row share K/V lines via L1/L2, and consecutive rows of a chunk
