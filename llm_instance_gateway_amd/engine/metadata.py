@@ -36,6 +36,9 @@ class BatchMeta:
     # cost 32x redundant arange/repeat kernels, see profiles/r01_*.md):
     prefill_tok_block_tables: Optional[torch.Tensor] = None  # [Tp, max_blocks]
     prefill_tok_context_lens: Optional[torch.Tensor] = None  # [Tp] int32
+    # 64-row query tiles for the MFMA flash-prefill kernel (GPU path).
+    prefill_tile_seq: Optional[torch.Tensor] = None  # [ntiles] int32
+    prefill_tile_qstart: Optional[torch.Tensor] = None  # [ntiles] int32
 
     # Multi-LoRA (SGMV): per-token adapter slot index, -1 = base model only.
     lora_slot_per_token: Optional[torch.Tensor] = None  # [T] int32

@@ -179,20 +179,36 @@ class ModelRunner:
         # Per-token prefill expansion, once per step (the attention layers of
         # every transformer block reuse it; see profiles/r01_bench_kernel_stats.md).
         if prefills:
-            ctx_tok: List[int] = []
-            n_per: List[int] = []
-            for ss in prefills:
-                c, n = ss.seq.num_computed_tokens, ss.num_tokens
-                ctx_tok.extend(range(c + 1, c + n + 1))
-                n_per.append(n)
-            meta.prefill_tok_context_lens = torch.tensor(
-                ctx_tok, dtype=torch.int32
-            ).to(dev, non_blocking=True)
-            meta.prefill_tok_block_tables = torch.repeat_interleave(
-                meta.prefill_block_tables,
-                torch.tensor(n_per, dtype=torch.int64, device=dev),
-                dim=0,
-            ).contiguous()
+            use_mfma = self.device.type == "cuda" and self.config.model.head_dim in (64, 128)
+            if use_mfma:
+                # 64-row query tiles for the MFMA flash-prefill kernel.
+                ts: List[int] = []
+                tq: List[int] = []
+                for si, ss in enumerate(prefills):
+                    for qb in range(0, ss.num_tokens, 64):
+                        ts.append(si)
+                        tq.append(qb)
+                meta.prefill_tile_seq = torch.tensor(ts, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
+                meta.prefill_tile_qstart = torch.tensor(tq, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
+            else:
+                ctx_tok: List[int] = []
+                n_per: List[int] = []
+                for ss in prefills:
+                    c, n = ss.seq.num_computed_tokens, ss.num_tokens
+                    ctx_tok.extend(range(c + 1, c + n + 1))
+                    n_per.append(n)
+                meta.prefill_tok_context_lens = torch.tensor(
+                    ctx_tok, dtype=torch.int32
+                ).to(dev, non_blocking=True)
+                meta.prefill_tok_block_tables = torch.repeat_interleave(
+                    meta.prefill_block_tables,
+                    torch.tensor(n_per, dtype=torch.int64, device=dev),
+                    dim=0,
+                ).contiguous()
 
         ids_t = torch.tensor(input_ids, dtype=torch.int64).to(dev, non_blocking=True)
         pos_t = torch.tensor(positions, dtype=torch.int64).to(dev, non_blocking=True)

@@ -87,6 +87,8 @@ class Attention(nn.Module):
                 self.scale,
                 tok_block_tables=meta.prefill_tok_block_tables,
                 tok_context_lens=meta.prefill_tok_context_lens,
+                tile_seq=meta.prefill_tile_seq,
+                tile_qstart=meta.prefill_tile_qstart,
             )
         if nd > 0:
             out[np_:] = ops.paged_attention_decode(

@@ -136,8 +136,29 @@ def paged_attention_prefill(
     scale: float,
     tok_block_tables: "Optional[torch.Tensor]" = None,
     tok_context_lens: "Optional[torch.Tensor]" = None,
+    tile_seq: "Optional[torch.Tensor]" = None,
+    tile_qstart: "Optional[torch.Tensor]" = None,
 ) -> torch.Tensor:
     if _use_native(q):
+        head_dim = q.shape[-1]
+        if head_dim in (64, 128):
+            # MFMA flash-prefill kernel over 64-row query tiles.
+            if tile_seq is None or tile_qstart is None:
+                qsl = query_start_locs.cpu().tolist()
+                ts, tq = [], []
+                for s in range(len(qsl) - 1):
+                    for qb in range(0, qsl[s + 1] - qsl[s], 64):
+                        ts.append(s)
+                        tq.append(qb)
+                dev = q.device
+                tile_seq = torch.tensor(ts, dtype=torch.int32, device=dev)
+                tile_qstart = torch.tensor(tq, dtype=torch.int32, device=dev)
+            out = torch.empty_like(q)
+            _EXT.paged_attention_prefill(
+                out, q, k_cache, v_cache, tile_seq, tile_qstart,
+                query_start_locs, context_lens, block_tables, scale, _stream()
+            )
+            return out
         if tok_block_tables is not None and tok_context_lens is not None:
             # per-token expansion precomputed once per step by the model runner
             return paged_attention_decode(

@@ -26,6 +26,12 @@ void lig_paged_attention_decode_launch(uint16_t*, const uint16_t*,
                                        const uint16_t*, const uint16_t*,
                                        const int*, const int*, float, int,
                                        int, int, int, int, int, hipStream_t);
+void lig_mfma_probe_launch(float*, const uint16_t*, const uint16_t*,
+                           hipStream_t);
+void lig_paged_attention_prefill_launch(
+    uint16_t*, const uint16_t*, const uint16_t*, const uint16_t*, const int*,
+    const int*, const int*, const int*, const int*, float, int, int, int,
+    int, int, int, hipStream_t);
 }
 
 namespace {
@@ -127,6 +133,34 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       (float)scale, N, Hq, Hkv, D, BS, max_blocks, as_stream(stream));
 }
 
+void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor k_cache, torch::Tensor v_cache,
+                             torch::Tensor tile_seq, torch::Tensor tile_qstart,
+                             torch::Tensor query_start_locs,
+                             torch::Tensor context_lens,
+                             torch::Tensor block_tables, double scale,
+                             int64_t stream) {
+  TORCH_CHECK(tile_seq.scalar_type() == torch::kInt32);
+  TORCH_CHECK(tile_qstart.scalar_type() == torch::kInt32);
+  TORCH_CHECK(query_start_locs.scalar_type() == torch::kInt32);
+  TORCH_CHECK(context_lens.scalar_type() == torch::kInt32);
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  const int ntiles = tile_seq.size(0);
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(D == 64 || D == 128, "prefill MFMA kernel needs head_dim 64/128");
+  if (ntiles == 0) return;
+  lig_paged_attention_prefill_launch(
+      bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
+      tile_seq.data_ptr<int>(), tile_qstart.data_ptr<int>(),
+      query_start_locs.data_ptr<int>(), context_lens.data_ptr<int>(),
+      block_tables.data_ptr<int>(), (float)scale, ntiles, Hq, Hkv, D, BS,
+      max_blocks, as_stream(stream));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -137,4 +171,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reshape_and_cache", &reshape_and_cache);
   m.def("silu_and_mul", &silu_and_mul);
   m.def("paged_attention_decode", &paged_attention_decode);
+  m.def("paged_attention_prefill", &paged_attention_prefill);
+  m.def("mfma_probe", [](torch::Tensor out, torch::Tensor a, torch::Tensor b,
+                         int64_t stream) {
+    TORCH_CHECK(out.scalar_type() == torch::kFloat32);
+    lig_mfma_probe_launch(out.data_ptr<float>(), bf16_cptr(a), bf16_cptr(b),
+                          as_stream(stream));
+  });
 }
