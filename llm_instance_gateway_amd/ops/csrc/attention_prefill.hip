@@ -59,7 +59,6 @@ __global__ __launch_bounds__(256) void paged_attention_prefill_kernel(
     uint16_t K[KVBLK][D];
     uint16_t VT[D][KVBLK];
     uint16_t P[4][16][KVBLK];
-    float L[4][16];
   };
   __shared__ Shared sm;
 
@@ -181,10 +180,17 @@ __global__ __launch_bounds__(256) void paged_attention_prefill_kernel(
     }
     psum = stride_reduce_sum<16, 4>(psum);
     l = l * alpha + psum;
+    // acc_o's rows follow the MFMA C layout (row = lg*4+r), while the
+    // per-lane softmax stats belong to row li — shuffle each row's alpha
+    // from the lane that owns it (alpha is replicated across lg groups, so
+    // lane (lg*4+r) holds row (lg*4+r)'s value).
+    float alpha_r[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) alpha_r[r] = __shfl(alpha, lg * 4 + r, 64);
 #pragma unroll
     for (int db = 0; db < DB; ++db) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) acc_o[db][r] *= alpha;
+      for (int r = 0; r < 4; ++r) acc_o[db][r] *= alpha_r[r];
     }
 
     // ---- P: C-layout -> A-fragment layout via per-wave LDS slab ------------
@@ -213,18 +219,22 @@ __global__ __launch_bounds__(256) void paged_attention_prefill_kernel(
   }
 
   // ---- epilogue -------------------------------------------------------------
-  if (lg == 0) sm.L[wave][li] = l;
-  // wave-private slab; same-wave ordering again suffices
+  // shuffle each output row's l from its owner lane FIRST (all lanes active —
+  // __shfl under divergence would read inactive lanes), then write.
+  float inv_l[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const float lr = __shfl(l, lg * 4 + r, 64);
+    inv_l[r] = lr > 0.f ? 1.0f / lr : 0.0f;
+  }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int orow = qt + wave * 16 + lg * 4 + r;  // C-layout row = M = qrow
     if (orow >= n) continue;
-    const float lr = sm.L[wave][lg * 4 + r];
-    const float inv_l = lr > 0.f ? 1.0f / lr : 0.0f;
     uint16_t* op = out + ((size_t)(lo + orow) * num_q_heads + h) * D;
 #pragma unroll
     for (int db = 0; db < DB; ++db) {
-      op[db * 16 + li] = f32_to_bf16(acc_o[db][r] * inv_l);
+      op[db * 16 + li] = f32_to_bf16(acc_o[db][r] * inv_l[r]);
     }
   }
 }
