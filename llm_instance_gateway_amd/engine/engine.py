@@ -83,6 +83,13 @@ class LLMEngine:
         out: SchedulerOutput = self.scheduler.schedule()
         if out.is_empty:
             return []
+        if self.adapter_manager is not None:
+            # make every scheduled adapter resident (async side-stream loads
+            # fenced with events) before the step touches the slot weights
+            self.adapter_manager.prepare(
+                [ss.seq.lora_name for ss in out.scheduled if ss.seq.lora_name]
+            )
+            self.runner.adapter_manager = self.adapter_manager
         sampled = self.runner.execute(out)
         for ss in out.scheduled:
             seq = ss.seq

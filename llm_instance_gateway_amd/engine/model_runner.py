@@ -210,6 +210,18 @@ class ModelRunner:
                     dim=0,
                 ).contiguous()
 
+        # multi-LoRA: per-token adapter slot ids (-1 = base model only)
+        mgr = getattr(self, "adapter_manager", None)
+        if mgr is not None and any(ss.seq.lora_name for ss in prefills + decodes):
+            slots: List[int] = []
+            for ss in prefills:
+                slots.extend([mgr.slot_of(ss.seq.lora_name)] * ss.num_tokens)
+            for ss in decodes:
+                slots.append(mgr.slot_of(ss.seq.lora_name))
+            meta.lora_slot_per_token = torch.tensor(slots, dtype=torch.int32).to(
+                dev, non_blocking=True
+            )
+
         ids_t = torch.tensor(input_ids, dtype=torch.int64).to(dev, non_blocking=True)
         pos_t = torch.tensor(positions, dtype=torch.int64).to(dev, non_blocking=True)
         return ids_t, pos_t, meta, sample_seqs, sample_idx, prefills + decodes

@@ -55,6 +55,9 @@ class Attention(nn.Module):
         self.o_proj = _linear(q_out, h, device, dtype)
         self.q_size = q_out
         self.kv_size = kv_out
+        # multi-LoRA hook (set by lora.manager.AdapterManager._attach)
+        self._lora = None
+        self._layer_idx = -1
 
     def forward(
         self,
@@ -65,6 +68,8 @@ class Attention(nn.Module):
         cos_sin_cache: torch.Tensor,
     ) -> torch.Tensor:
         qkv = self.qkv_proj(hidden)
+        if self._lora is not None:
+            self._lora.apply(self._layer_idx, "qkv", hidden, qkv, meta)
         q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
         q = q.contiguous()
         k = k.contiguous()
@@ -99,7 +104,10 @@ class Attention(nn.Module):
                 meta.decode_context_lens,
                 self.scale,
             )
-        o = self.o_proj(out.view(-1, self.num_heads * self.head_dim))
+        attn_flat = out.view(-1, self.num_heads * self.head_dim)
+        o = self.o_proj(attn_flat)
+        if self._lora is not None:
+            self._lora.apply(self._layer_idx, "o", attn_flat, o, meta)
         if self.tp is not None:
             o = self.tp.all_reduce(o)
         return o
@@ -116,10 +124,17 @@ class MLP(nn.Module):
         # gate and up fused into one GEMM (silu_and_mul splits them)
         self.gate_up_proj = _linear(h, 2 * inter, device, dtype)
         self.down_proj = _linear(inter, h, device, dtype)
+        self._lora = None
+        self._layer_idx = -1
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = ops.silu_and_mul(self.gate_up_proj(x))
-        y = self.down_proj(y)
+    def forward(self, x: torch.Tensor, meta=None) -> torch.Tensor:
+        gu = self.gate_up_proj(x)
+        if self._lora is not None:
+            self._lora.apply(self._layer_idx, "gate_up", x, gu, meta)
+        act = ops.silu_and_mul(gu)
+        y = self.down_proj(act)
+        if self._lora is not None:
+            self._lora.apply(self._layer_idx, "down", act, y, meta)
         if self.tp is not None:
             y = self.tp.all_reduce(y)
         return y
@@ -148,7 +163,7 @@ class DecoderLayer(nn.Module):
         h, residual = ops.fused_add_rms_norm(
             h, residual, self.post_attention_layernorm, self.eps
         )
-        h = self.mlp(h)
+        h = self.mlp(h, meta)
         return h, residual
 
 

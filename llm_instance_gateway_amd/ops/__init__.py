@@ -197,3 +197,20 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
 
 
 build_cos_sin_cache = torch_ref.build_cos_sin_cache
+
+
+def sgmv_shrink(x, a_stack, slots):
+    if _use_native(x):
+        T = x.shape[0]
+        R = a_stack.shape[1]
+        u = torch.zeros(T, R, dtype=x.dtype, device=x.device)
+        _EXT.sgmv_shrink(u, x, a_stack, slots, _stream())
+        return u
+    return torch_ref.sgmv_shrink(x, a_stack, slots)
+
+
+def sgmv_expand(u, b_stack, slots, y, scaling=1.0):
+    if _use_native(u):
+        _EXT.sgmv_expand(y, u, b_stack, slots, float(scaling), _stream())
+        return
+    torch_ref.sgmv_expand(u, b_stack, slots, y, scaling)

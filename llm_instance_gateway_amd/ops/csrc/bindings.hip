@@ -28,6 +28,10 @@ void lig_paged_attention_decode_launch(uint16_t*, const uint16_t*,
                                        int, int, int, int, int, hipStream_t);
 void lig_mfma_probe_launch(float*, const uint16_t*, const uint16_t*,
                            hipStream_t);
+void lig_sgmv_shrink_launch(uint16_t*, const uint16_t*, const uint16_t*,
+                            const int*, int, int, int, hipStream_t);
+void lig_sgmv_expand_launch(uint16_t*, const uint16_t*, const uint16_t*,
+                            const int*, float, int, int, int, hipStream_t);
 void lig_paged_attention_prefill_launch(
     uint16_t*, const uint16_t*, const uint16_t*, const uint16_t*, const int*,
     const int*, const int*, const int*, const int*, float, int, int, int,
@@ -161,6 +165,33 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
       max_blocks, as_stream(stream));
 }
 
+void sgmv_shrink(torch::Tensor u, torch::Tensor x, torch::Tensor a_stack,
+                 torch::Tensor slots, int64_t stream) {
+  TORCH_CHECK(slots.scalar_type() == torch::kInt32);
+  const int T = x.size(0);
+  const int H = x.size(1);
+  const int R = a_stack.size(1);
+  TORCH_CHECK(a_stack.size(2) == H, "A stack in-features mismatch");
+  TORCH_CHECK(H % 8 == 0 && R <= 64);
+  if (T == 0) return;
+  lig_sgmv_shrink_launch(bf16_ptr(u), bf16_cptr(x), bf16_cptr(a_stack),
+                         slots.data_ptr<int>(), T, R, H, as_stream(stream));
+}
+
+void sgmv_expand(torch::Tensor y, torch::Tensor u, torch::Tensor b_stack,
+                 torch::Tensor slots, double scaling, int64_t stream) {
+  TORCH_CHECK(slots.scalar_type() == torch::kInt32);
+  const int T = u.size(0);
+  const int R = u.size(1);
+  const int O = y.size(1);
+  TORCH_CHECK(b_stack.size(1) == O && b_stack.size(2) == R, "B stack mismatch");
+  TORCH_CHECK(O % 8 == 0 && R <= 64);
+  if (T == 0) return;
+  lig_sgmv_expand_launch(bf16_ptr(y), bf16_cptr(u), bf16_cptr(b_stack),
+                         slots.data_ptr<int>(), (float)scaling, T, R, O,
+                         as_stream(stream));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -172,6 +203,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_and_mul", &silu_and_mul);
   m.def("paged_attention_decode", &paged_attention_decode);
   m.def("paged_attention_prefill", &paged_attention_prefill);
+  m.def("sgmv_shrink", &sgmv_shrink);
+  m.def("sgmv_expand", &sgmv_expand);
   m.def("mfma_probe", [](torch::Tensor out, torch::Tensor a, torch::Tensor b,
                          int64_t stream) {
     TORCH_CHECK(out.scalar_type() == torch::kFloat32);

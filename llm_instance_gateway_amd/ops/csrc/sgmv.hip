@@ -1,0 +1,106 @@
+// SGMV (segmented grouped matrix-vector) kernels for batched multi-LoRA,
+// bf16, gfx950.
+//
+// The batched multi-LoRA capability the reference routes around but delegates
+// to vLLM's punica kernels (SURVEY.md §2.3 "Batched multi-LoRA serving"):
+// every token in a mixed batch may use a different adapter; slot_per_token
+// maps token -> adapter slot (-1 = base model only).
+//
+//   shrink:  u[t, :R]  = A[slot_t] (R x H)  · x[t]      (R = lora rank, <=64)
+//   expand:  y[t, :O] += s * B[slot_t] (O x R) · u[t]
+//
+// Both are memory-bound on the adapter weights; tokens sharing an adapter hit
+// the same A/B rows in L2 (256 MiB LLC holds tens of rank-16 adapter pairs).
+//
+// shrink: one workgroup per token; wave w handles rank rows w, w+4, ... —
+//   each lane covers H/64 dims (vectorized u16x8), wave-reduce per row.
+// expand: grid-stride over (token, 16-wide output chunks); u[t] (R values)
+//   is read via L1; each thread accumulates 16 outputs from B rows.
+
+#include "common.h"
+
+__global__ void sgmv_shrink_kernel(
+    uint16_t* __restrict__ u,        // [T, R] bf16 (zero-filled by caller)
+    const uint16_t* __restrict__ x,  // [T, H]
+    const uint16_t* __restrict__ a,  // [S, R, H]
+    const int* __restrict__ slots,   // [T]
+    const int T, const int R, const int H) {
+  const int t = blockIdx.x;
+  const int s = slots[t];
+  if (s < 0) return;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int nwaves = blockDim.x >> 6;
+  const uint16_t* xrow = x + (size_t)t * H;
+  const int chunks = H / 8;  // 8 bf16 per lane-step
+
+  for (int r = wave; r < R; r += nwaves) {
+    const uint16_t* arow = a + ((size_t)s * R + r) * H;
+    float acc = 0.0f;
+    for (int c = lane; c < chunks; c += 64) {
+      f32x8 xv = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(xrow + c * 8));
+      f32x8 av = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(arow + c * 8));
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc += xv[e] * av[e];
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) u[(size_t)t * R + r] = f32_to_bf16(acc);
+  }
+}
+
+__global__ void sgmv_expand_kernel(
+    uint16_t* __restrict__ y,        // [T, O] accumulated in place
+    const uint16_t* __restrict__ u,  // [T, R]
+    const uint16_t* __restrict__ b,  // [S, O, R]
+    const int* __restrict__ slots,   // [T]
+    const float scaling,
+    const int T, const int R, const int O) {
+  // thread handles 8 consecutive outputs of one token
+  const long total = (long)T * (O / 8);
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += gridDim.x * (long)blockDim.x) {
+    const int oc = idx % (O / 8);
+    const long t = idx / (O / 8);
+    const int s = slots[t];
+    if (s < 0) continue;
+    // u[t] is tiny (R<=64) and L1-hot across the token's O/8 threads
+    float uv[64];
+    for (int r = 0; r < R; ++r) uv[r] = bf16_to_f32(u[(size_t)t * R + r]);
+    const uint16_t* brow = b + ((size_t)s * O + oc * 8) * R;
+    f32x8 acc;
+    u16x8 yv = *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = bf16_to_f32(yv[e]);
+    for (int e = 0; e < 8; ++e) {
+      const uint16_t* br = brow + (size_t)e * R;
+      float a = 0.0f;
+      for (int r = 0; r < R; ++r) a += bf16_to_f32(br[r]) * uv[r];
+      acc[e] += a * scaling;
+    }
+    *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8) = f32x8_to_bf16x8(acc);
+  }
+}
+
+extern "C" {
+
+void lig_sgmv_shrink_launch(uint16_t* u, const uint16_t* x, const uint16_t* a,
+                            const int* slots, int T, int R, int H,
+                            hipStream_t stream) {
+  dim3 grid(T);
+  dim3 block(256);
+  hipLaunchKernelGGL(sgmv_shrink_kernel, grid, block, 0, stream, u, x, a,
+                     slots, T, R, H);
+}
+
+void lig_sgmv_expand_launch(uint16_t* y, const uint16_t* u, const uint16_t* b,
+                            const int* slots, float scaling, int T, int R,
+                            int O, hipStream_t stream) {
+  const long total = (long)T * (O / 8);
+  int grid = (int)((total + 255) / 256);
+  if (grid > 65535) grid = 65535;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(sgmv_expand_kernel, dim3(grid), dim3(256), 0, stream, y,
+                     u, b, slots, scaling, T, R, O);
+}
+
+}  // extern "C"

@@ -182,3 +182,32 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     a = x[..., :d].float()
     b = x[..., d:].float()
     return (a * torch.sigmoid(a) * b).to(x.dtype)
+
+
+def sgmv_shrink(
+    x: torch.Tensor,  # [T, H]
+    a_stack: torch.Tensor,  # [S, R, H]
+    slots: torch.Tensor,  # [T] int32, -1 = no adapter
+) -> torch.Tensor:
+    """u[t] = A[slot_t] @ x[t]; zero rows for slot -1."""
+    T = x.shape[0]
+    R = a_stack.shape[1]
+    u = torch.zeros(T, R, dtype=torch.float32, device=x.device)
+    for t in range(T):
+        s = int(slots[t])
+        if s >= 0:
+            u[t] = a_stack[s].float() @ x[t].float()
+    return u.to(x.dtype)
+
+
+def sgmv_expand(
+    u: torch.Tensor,  # [T, R]
+    b_stack: torch.Tensor,  # [S, O, R]
+    slots: torch.Tensor,  # [T]
+    y: torch.Tensor,  # [T, O] accumulated in-place
+    scaling: float = 1.0,
+) -> None:
+    for t in range(u.shape[0]):
+        s = int(slots[t])
+        if s >= 0:
+            y[t] += (b_stack[s].float() @ u[t].float() * scaling).to(y.dtype)

@@ -164,3 +164,58 @@ def test_engine_e2e_gpu_matches_cpu_greedy():
     )
     total = sum(len(cs.output_token_ids) for cs in cseqs)
     assert agree >= total * 0.7, f"only {agree}/{total} tokens agree"
+
+
+@pytest.mark.parametrize("T,H,R,O", [(64, 4096, 16, 6144), (3, 64, 8, 128)])
+def test_sgmv_kernels(T, H, R, O):
+    torch.manual_seed(9)
+    S = 4
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    a = torch.randn(S, R, H, dtype=torch.bfloat16, device=DEV) * 0.05
+    b = torch.randn(S, O, R, dtype=torch.bfloat16, device=DEV) * 0.05
+    slots = torch.randint(-1, S, (T,), dtype=torch.int32, device=DEV)
+    slots[0] = -1
+    u_got = ops.sgmv_shrink(x, a, slots)
+    u_ref = torch_ref.sgmv_shrink(x, a, slots)
+    close(u_got, u_ref)
+    y_got = torch.randn(T, O, dtype=torch.bfloat16, device=DEV)
+    y_ref = y_got.clone()
+    ops.sgmv_expand(u_ref.to(torch.bfloat16), b, slots, y_got, 1.0)
+    torch_ref.sgmv_expand(u_ref, b, slots, y_ref, 1.0)
+    close(y_got, y_ref)
+
+
+def test_lora_engine_gpu_matches_cpu():
+    from llm_instance_gateway_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from llm_instance_gateway_amd.lora.manager import AdapterManager
+    from llm_instance_gateway_amd.models.config import get_model_config
+
+    def build(device):
+        torch.manual_seed(3)
+        cfg = EngineConfig(
+            model=get_model_config("tiny"),
+            num_kv_blocks=64,
+            max_num_seqs=8,
+            max_num_batched_tokens=64,
+            max_model_len=256,
+            max_loras=2,
+        )
+        eng = LLMEngine(cfg, device=device)
+        mgr = AdapterManager(eng, max_slots=2, rank=4)
+        mgr.register_random("s-a", seed=101)
+        return eng
+
+    cpu = build("cpu")
+    gpu = build(DEV)
+    gpu.runner.model.load_state_dict(
+        {k: v.to(DEV) for k, v in cpu.runner.model.state_dict().items()}
+    )
+    g = torch.Generator().manual_seed(5)
+    prompt = torch.randint(1, 500, (20,), generator=g).tolist()
+    sp = SamplingParams(max_new_tokens=6)
+    a = cpu.add_request(prompt, sp, lora_name="s-a")
+    cpu.generate_all()
+    b = gpu.add_request(prompt, sp, lora_name="s-a")
+    gpu.generate_all()
+    agree = sum(int(x == y) for x, y in zip(a.output_token_ids, b.output_token_ids))
+    assert agree >= len(a.output_token_ids) * 0.65
