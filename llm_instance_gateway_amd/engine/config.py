@@ -38,7 +38,7 @@ class EngineConfig:
     # Engine identity.
     engine_id: str = "engine-0"
     gpu_id: int = 0
-    enforce_eager: bool = True  # hipGraph capture of the decode step when False
+    enforce_eager: bool = False  # False: decode-only steps replay hipGraphs
 
     def __post_init__(self) -> None:
         if self.block_size <= 0:

@@ -82,7 +82,8 @@ class ModelRunner:
                 n, config.num_kv_blocks if config.num_kv_blocks > 0 else n
             )
         kvh = mc.num_kv_heads // tp_size
-        shape = (config.num_kv_blocks, kvh, config.block_size, mc.head_dim)
+        # +1 scratch block: hipGraph decode padding rows write their K/V there
+        shape = (config.num_kv_blocks + 1, kvh, config.block_size, mc.head_dim)
         self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = [
             (
                 torch.zeros(shape, dtype=dtype, device=self.device),
@@ -95,6 +96,11 @@ class ModelRunner:
             config.num_kv_blocks,
             config.block_size,
             2 * mc.num_layers * shape[0] * kvh * config.block_size * mc.head_dim * 2 / 2**30,
+        )
+        self.adapter_manager = None
+        self.graph_runner = None
+        self._use_graphs = (
+            self.device.type == "cuda" and not config.enforce_eager
         )
 
     # -- batch construction ---------------------------------------------------
@@ -233,6 +239,10 @@ class ModelRunner:
         num_computed_tokens for every scheduled sequence."""
         if out.is_empty:
             return {}
+        if self._use_graphs and all(ss.num_tokens == 1 for ss in out.scheduled):
+            res = self._execute_graph(out)
+            if res is not None:
+                return res
         ids, pos, meta, sample_seqs, sample_idx, ordered = self._build_batch(out)
         hidden = self.model(ids, pos, self.kv_caches, meta)
         sampled: Dict[int, int] = {}
@@ -245,3 +255,23 @@ class ModelRunner:
         for ss in ordered:
             ss.seq.num_computed_tokens += ss.num_tokens
         return sampled
+
+    def _execute_graph(self, out: SchedulerOutput) -> Optional[Dict[int, int]]:
+        """Decode-only fast path: one hipGraph replay for the whole step."""
+        from .graph_runner import DecodeGraphRunner
+
+        decodes = out.scheduled
+        want_lora = self.adapter_manager is not None
+        if self.graph_runner is None or self.graph_runner.with_lora != want_lora:
+            self.graph_runner = DecodeGraphRunner(self)
+            self.graph_runner.capture(with_lora=want_lora)
+        gr = self.graph_runner
+        if gr.pick_size(len(decodes)) is None:
+            return None  # larger than any captured bucket: eager fallback
+        slot_of = self.adapter_manager.slot_of if want_lora else None
+        logits = gr.run(decodes, slot_of)
+        seqs = [ss.seq for ss in decodes]
+        tokens = sample_tokens(logits.float(), seqs)
+        for ss in decodes:
+            ss.seq.num_computed_tokens += 1
+        return {seq.seq_id: tok for seq, tok in zip(seqs, tokens)}
