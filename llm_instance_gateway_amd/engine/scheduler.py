@@ -127,12 +127,22 @@ class EngineScheduler:
 
     # -- the step builder -------------------------------------------------------
     def schedule(self) -> SchedulerOutput:
-        out = SchedulerOutput()
+        """One step is EITHER all-prefill-chunks OR all-decodes (the
+        reference simulation's prefill_or_decode loop,
+        continous_batching.py:173-191, and what lets decode steps replay a
+        captured hipGraph — see engine/graph_runner.py).  Prefill work is
+        prioritised (TTFT); decodes run when no prefill is admittable."""
+        out = self._schedule_prefills()
+        if not out.is_empty:
+            return out
+        return self._schedule_decodes(out)
+
+    def _schedule_decodes(self, out: SchedulerOutput) -> SchedulerOutput:
         cfg = self.config
         bm = self.block_manager
         token_budget = cfg.max_num_batched_tokens
 
-        # 1) Decodes: every running seq past prefill needs one more slot
+        # Decodes: every running seq past prefill needs one more slot
         #    (the token at position total_len-1 gets its KV cached this step).
         #    Oldest sequences are served first; when blocks run out, the
         #    newest not-yet-scheduled sequence is preempted.
@@ -158,8 +168,15 @@ class EngineScheduler:
                 scheduled_ids.add(seq.seq_id)
                 token_budget -= 1
                 continue
+        return out
 
-        # 2) Continue chunked prefills already running.
+    def _schedule_prefills(self) -> SchedulerOutput:
+        out = SchedulerOutput()
+        cfg = self.config
+        bm = self.block_manager
+        token_budget = cfg.max_num_batched_tokens
+
+        # 1) Continue chunked prefills already running.
         for seq in list(self.running):
             if not seq.in_prefill or token_budget <= 0:
                 continue
@@ -173,7 +190,7 @@ class EngineScheduler:
             out.scheduled.append(ScheduledSeq(seq, chunk))
             token_budget -= chunk
 
-        # 3) Admit waiting sequences (FIFO) under caps.
+        # 2) Admit waiting sequences (FIFO) under caps.
         while (
             self.waiting
             and token_budget > 0
