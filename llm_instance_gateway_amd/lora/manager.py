@@ -270,5 +270,18 @@ class AdapterManager:
         from .. import ops
 
         key = (layer_idx, module)
-        u = ops.sgmv_shrink(x, self.a_stacks[key], slots)
-        ops.sgmv_expand(u, self.b_stacks[key], slots, y, 1.0)
+        if ops.native_available() and x.is_cuda:
+            T = x.shape[0]
+            ws = getattr(self, "_u_ws", None)
+            if ws is None or ws.shape[0] < T:
+                ws = torch.zeros(
+                    max(T, 4096), self.rank, dtype=self.dtype, device=self.device
+                )
+                self._u_ws = ws
+            ops._EXT.sgmv_apply(
+                y, x, self.a_stacks[key], self.b_stacks[key], ws, slots, 1.0,
+                ops._stream(),
+            )
+        else:
+            u = ops.sgmv_shrink(x, self.a_stacks[key], slots)
+            ops.sgmv_expand(u, self.b_stacks[key], slots, y, 1.0)

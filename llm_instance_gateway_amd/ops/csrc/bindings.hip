@@ -178,6 +178,25 @@ void sgmv_shrink(torch::Tensor u, torch::Tensor x, torch::Tensor a_stack,
                          slots.data_ptr<int>(), T, R, H, as_stream(stream));
 }
 
+// one call = shrink + expand (halves Python dispatch; u is a caller-owned
+// workspace so no per-call allocation)
+void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
+                torch::Tensor b_stack, torch::Tensor u_ws,
+                torch::Tensor slots, double scaling, int64_t stream) {
+  TORCH_CHECK(slots.scalar_type() == torch::kInt32);
+  const int T = x.size(0);
+  const int H = x.size(1);
+  const int R = a_stack.size(1);
+  const int O = y.size(1);
+  TORCH_CHECK(u_ws.size(0) >= T && u_ws.size(1) == R, "u workspace too small");
+  if (T == 0) return;
+  hipStream_t st = as_stream(stream);
+  lig_sgmv_shrink_launch(bf16_ptr(u_ws), bf16_cptr(x), bf16_cptr(a_stack),
+                         slots.data_ptr<int>(), T, R, H, st);
+  lig_sgmv_expand_launch(bf16_ptr(y), bf16_cptr(u_ws), bf16_cptr(b_stack),
+                         slots.data_ptr<int>(), (float)scaling, T, R, O, st);
+}
+
 void sgmv_expand(torch::Tensor y, torch::Tensor u, torch::Tensor b_stack,
                  torch::Tensor slots, double scaling, int64_t stream) {
   TORCH_CHECK(slots.scalar_type() == torch::kInt32);
@@ -204,6 +223,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_decode", &paged_attention_decode);
   m.def("paged_attention_prefill", &paged_attention_prefill);
   m.def("sgmv_shrink", &sgmv_shrink);
+  m.def("sgmv_apply", &sgmv_apply);
   m.def("sgmv_expand", &sgmv_expand);
   m.def("mfma_probe", [](torch::Tensor out, torch::Tensor a, torch::Tensor b,
                          int64_t stream) {
