@@ -189,6 +189,7 @@ void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
   const int R = a_stack.size(1);
   const int O = y.size(1);
   TORCH_CHECK(u_ws.size(0) >= T && u_ws.size(1) == R, "u workspace too small");
+  TORCH_CHECK(R % 8 == 0 && R <= 64, "rank must be 8/16/32/64");
   if (T == 0) return;
   hipStream_t st = as_stream(stream);
   lig_sgmv_shrink_launch(bf16_ptr(u_ws), bf16_cptr(x), bf16_cptr(a_stack),
@@ -204,7 +205,7 @@ void sgmv_expand(torch::Tensor y, torch::Tensor u, torch::Tensor b_stack,
   const int R = u.size(1);
   const int O = y.size(1);
   TORCH_CHECK(b_stack.size(1) == O && b_stack.size(2) == R, "B stack mismatch");
-  TORCH_CHECK(O % 8 == 0 && R <= 64);
+  TORCH_CHECK(O % 8 == 0 && R % 8 == 0 && R <= 64, "rank must be 8/16/32/64");
   if (T == 0) return;
   lig_sgmv_expand_launch(bf16_ptr(y), bf16_cptr(u), bf16_cptr(b_stack),
                          slots.data_ptr<int>(), (float)scaling, T, R, O,

@@ -48,13 +48,17 @@ __global__ void sgmv_shrink_kernel(
   }
 }
 
+// R is a template parameter (multiple of 8) so every B-row and u read is a
+// full 16 B u16x8 vector load — the scalar-2B-load version of this kernel
+// was 73% of total step time (guide common-mistake #2: never scalar bf16).
+template <int R>
 __global__ void sgmv_expand_kernel(
     uint16_t* __restrict__ y,        // [T, O] accumulated in place
     const uint16_t* __restrict__ u,  // [T, R]
     const uint16_t* __restrict__ b,  // [S, O, R]
     const int* __restrict__ slots,   // [T]
     const float scaling,
-    const int T, const int R, const int O) {
+    const int T, const int O) {
   // thread handles 8 consecutive outputs of one token
   const long total = (long)T * (O / 8);
   for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
@@ -63,18 +67,30 @@ __global__ void sgmv_expand_kernel(
     const long t = idx / (O / 8);
     const int s = slots[t];
     if (s < 0) continue;
-    // u[t] is tiny (R<=64) and L1-hot across the token's O/8 threads
-    float uv[64];
-    for (int r = 0; r < R; ++r) uv[r] = bf16_to_f32(u[(size_t)t * R + r]);
+    // u[t]: R/8 vector loads; L1-hot across the token's O/8 threads
+    float uv[R];
+#pragma unroll
+    for (int rc = 0; rc < R / 8; ++rc) {
+      f32x8 v = bf16x8_to_f32x8(
+          *reinterpret_cast<const u16x8*>(u + (size_t)t * R + rc * 8));
+#pragma unroll
+      for (int e = 0; e < 8; ++e) uv[rc * 8 + e] = v[e];
+    }
     const uint16_t* brow = b + ((size_t)s * O + oc * 8) * R;
     f32x8 acc;
     u16x8 yv = *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8);
 #pragma unroll
     for (int e = 0; e < 8; ++e) acc[e] = bf16_to_f32(yv[e]);
+#pragma unroll
     for (int e = 0; e < 8; ++e) {
-      const uint16_t* br = brow + (size_t)e * R;
       float a = 0.0f;
-      for (int r = 0; r < R; ++r) a += bf16_to_f32(br[r]) * uv[r];
+#pragma unroll
+      for (int rc = 0; rc < R / 8; ++rc) {
+        f32x8 bv = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(
+            brow + (size_t)e * R + rc * 8));
+#pragma unroll
+        for (int q = 0; q < 8; ++q) a += bv[q] * uv[rc * 8 + q];
+      }
       acc[e] += a * scaling;
     }
     *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8) = f32x8_to_bf16x8(acc);
@@ -99,8 +115,17 @@ void lig_sgmv_expand_launch(uint16_t* y, const uint16_t* u, const uint16_t* b,
   int grid = (int)((total + 255) / 256);
   if (grid > 65535) grid = 65535;
   if (grid < 1) grid = 1;
-  hipLaunchKernelGGL(sgmv_expand_kernel, dim3(grid), dim3(256), 0, stream, y,
-                     u, b, slots, scaling, T, R, O);
+#define LAUNCH_E(RR)                                                       \
+  hipLaunchKernelGGL((sgmv_expand_kernel<RR>), dim3(grid), dim3(256), 0,   \
+                     stream, y, u, b, slots, scaling, T, O)
+  switch (R) {
+    case 8: LAUNCH_E(8); break;
+    case 16: LAUNCH_E(16); break;
+    case 32: LAUNCH_E(32); break;
+    case 64: LAUNCH_E(64); break;
+    default: break;  // caller validates R % 8 == 0 && R <= 64
+  }
+#undef LAUNCH_E
 }
 
 }  // extern "C"

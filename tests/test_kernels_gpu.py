@@ -201,7 +201,7 @@ def test_lora_engine_gpu_matches_cpu():
             max_loras=2,
         )
         eng = LLMEngine(cfg, device=device)
-        mgr = AdapterManager(eng, max_slots=2, rank=4)
+        mgr = AdapterManager(eng, max_slots=2, rank=8)
         mgr.register_random("s-a", seed=101)
         return eng
 
