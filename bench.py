@@ -69,7 +69,13 @@ def main() -> None:
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--concurrency", type=int, default=128)
     ap.add_argument("--device", default=None, help="override (e.g. cpu for dev)")
-    ap.add_argument("--lora", action="store_true", help="activate LoRA adapters (SGMV path)")
+    ap.add_argument(
+        "--lora",
+        action=argparse.BooleanOptionalAction,
+        default=True,
+        help="LoRA adapters on the SGMV path (default ON: BASELINE's metric "
+        "is Llama-3-8B + 32 LoRAs; --no-lora for the base-model A/B)",
+    )
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))

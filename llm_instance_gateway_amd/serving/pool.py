@@ -1,0 +1,602 @@
+"""Engine pool: N per-GPU engine worker processes behind one router.
+
+MI355X-native replacement for the reference's pod fleet + Envoy dispatch
+(SURVEY.md §7): instead of K8s pods scraped over HTTP every 50 ms
+(`pkg/ext-proc/backend/provider.go:134-179`) and routed via a `target-pod`
+header (`pkg/manifests/patch_policy.yaml:24-36`), the engines are worker
+processes on the same node — one per GPU, required because each engine's
+host-side continuous-batching loop must not share a GIL — publishing their
+gauges into a **shared-memory array the router reads with zero staleness**,
+and receiving requests as direct IPC messages.
+
+Per-engine shared gauge slots (doubles, written lock-free by the worker after
+every scheduler iteration, read on every route):
+
+    [heartbeat_ts, step_count, waiting, running, kv_usage, kv_capacity,
+     max_loras, draining]
+
+The variable-size resident-adapter set travels as an event message only when
+it changes (rare: adapter load/evict), cached parent-side.
+
+Failure semantics (reference parity): a worker that stops
+heartbeating is marked unhealthy and drops out of the routing set (the
+reference's EndpointSlice `Ready` condition, `endpointslice_reconciler.go:
+107-111`); `restart_engine` re-spawns it and replays the adapter registry
+(the reference's pod restart + sidecar reconcile).
+"""
+
+from __future__ import annotations
+
+import logging
+import multiprocessing as mp
+import os
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional, Sequence
+
+from ..router.datastore import Datastore
+from ..router.scheduler import ResourceExhausted, Scheduler
+from ..router.types import Engine, EngineMetrics, EngineState, LLMRequest
+
+log = logging.getLogger(__name__)
+
+# shared gauge slot layout
+_SLOT_HEARTBEAT = 0
+_SLOT_STEP = 1
+_SLOT_WAITING = 2
+_SLOT_RUNNING = 3
+_SLOT_KV_USAGE = 4
+_SLOT_KV_CAP = 5
+_SLOT_MAX_LORAS = 6
+_SLOT_DRAINING = 7
+SLOTS_PER_ENGINE = 8
+
+HEARTBEAT_STALE_S = 30.0
+
+
+@dataclass
+class PoolRequest:
+    """One request as handed to an engine worker."""
+
+    request_id: str
+    prompt_token_ids: List[int]
+    max_new_tokens: int = 128
+    temperature: float = 0.0
+    top_p: float = 1.0
+    lora_name: str = ""
+    critical: bool = False
+    stream: bool = False
+    arrival_time: float = field(default_factory=time.monotonic)
+
+
+@dataclass
+class FinishedRequest:
+    request_id: str
+    engine_name: str
+    prompt_len: int
+    output_token_ids: List[int]
+    ttft: Optional[float]
+    tpot: Optional[float]
+    e2e: float
+    finish_reason: str = "length"
+
+
+class RequestTicket:
+    """Parent-side handle for one in-flight request (thread-safe; the async
+    front door bridges it onto the event loop)."""
+
+    def __init__(self, request_id: str, engine_name: str, stream: bool):
+        self.request_id = request_id
+        self.engine_name = engine_name
+        self.stream = stream
+        self.tokens: "queue.Queue[Optional[int]]" = queue.Queue()
+        self.done = threading.Event()
+        self.result: Optional[FinishedRequest] = None
+        self.error: Optional[str] = None
+
+    def wait(self, timeout: Optional[float] = None) -> FinishedRequest:
+        if not self.done.wait(timeout):
+            raise TimeoutError(f"request {self.request_id} timed out")
+        if self.error:
+            raise RuntimeError(self.error)
+        assert self.result is not None
+        return self.result
+
+
+def _worker_main(
+    conn,
+    shared,
+    slot_base: int,
+    engine_name: str,
+    model_name: str,
+    device: str,
+    engine_kwargs: Dict[str, Any],
+    lora_rank: int,
+) -> None:
+    """Engine worker process body: one LLMEngine, one GPU, one event loop."""
+    import torch
+
+    from ..engine import EngineConfig, LLMEngine, SamplingParams
+    from ..lora.manager import AdapterManager
+    from ..models.config import get_model_config
+
+    if device.startswith("cuda"):
+        torch.cuda.set_device(device)
+    mc = get_model_config(model_name)
+    cfg = EngineConfig(model=mc, engine_id=engine_name, **engine_kwargs)
+    torch.manual_seed(1234)  # same base weights on every engine of the pool
+    eng = LLMEngine(cfg, device=device)
+    mgr: Optional[AdapterManager] = None
+    if cfg.max_loras > 0:
+        mgr = AdapterManager(eng, max_slots=cfg.max_loras, rank=lora_rank)
+
+    seq_by_id: Dict[str, Any] = {}
+    stream_ids: set = set()
+    token_buf: List = []
+    draining = False
+
+    def publish_gauges() -> None:
+        m = eng.get_metrics()
+        b = slot_base
+        shared[b + _SLOT_HEARTBEAT] = time.time()
+        shared[b + _SLOT_STEP] = float(eng._step_count)
+        shared[b + _SLOT_WAITING] = float(m.waiting_queue_size)
+        shared[b + _SLOT_RUNNING] = float(m.running_queue_size)
+        shared[b + _SLOT_KV_USAGE] = m.kv_cache_usage_percent
+        shared[b + _SLOT_KV_CAP] = float(m.kv_cache_max_token_capacity)
+        shared[b + _SLOT_MAX_LORAS] = float(m.max_active_models)
+        shared[b + _SLOT_DRAINING] = 1.0 if draining else 0.0
+
+    def publish_adapters() -> None:
+        if mgr is not None:
+            conn.send(("adapters", mgr.resident_names(), mgr.max_slots))
+
+    def on_token(seq, tok):
+        if seq.request_id in stream_ids:
+            token_buf.append((seq.request_id, tok))
+
+    eng.on_token = on_token
+
+    publish_gauges()
+    conn.send(("ready",))
+    prev_resident: List[str] = []
+
+    while True:
+        # drain all pending commands, then (maybe) step
+        timeout = 0.0 if eng.has_work() else 0.002
+        try:
+            while conn.poll(timeout):
+                msg = conn.recv()
+                op = msg[0]
+                if op == "add":
+                    r: PoolRequest = msg[1]
+                    if r.stream:
+                        stream_ids.add(r.request_id)
+                    try:
+                        seq = eng.add_request(
+                            r.prompt_token_ids,
+                            SamplingParams(
+                                max_new_tokens=r.max_new_tokens,
+                                temperature=r.temperature,
+                                top_p=r.top_p,
+                            ),
+                            lora_name=r.lora_name,
+                            request_id=r.request_id,
+                            critical=r.critical,
+                            arrival_time=r.arrival_time,
+                        )
+                        seq_by_id[r.request_id] = seq
+                    except Exception as e:  # e.g. prompt > max_model_len
+                        conn.send(("error", r.request_id, str(e)))
+                elif op == "abort":
+                    eng.abort_request(msg[1])
+                elif op == "load_adapter":
+                    _, name, source, rank = msg
+                    try:
+                        if mgr is None:
+                            raise RuntimeError("engine has no adapter slots")
+                        if isinstance(source, dict):
+                            mgr.register_weights(name, source, rank=rank)
+                        elif isinstance(source, str) and os.path.exists(source):
+                            weights = torch.load(source, weights_only=True)
+                            mgr.register_weights(name, weights, rank=rank)
+                        else:
+                            mgr.register_random(
+                                name, seed=abs(hash(name)) % (2**31), rank=rank
+                            )
+                        mgr.prepare([name])
+                        conn.send(("adapter_loaded", name, None))
+                    except Exception as e:
+                        conn.send(("adapter_loaded", name, str(e)))
+                    publish_adapters()
+                elif op == "unload_adapter":
+                    name = msg[1]
+                    ok = mgr.unregister(name) if mgr is not None else False
+                    conn.send(("adapter_unloaded", name, ok))
+                    publish_adapters()
+                elif op == "drain":
+                    draining = bool(msg[1])
+                elif op == "ping":
+                    conn.send(("pong", msg[1]))
+                elif op == "stop":
+                    return
+                timeout = 0.0
+        except (EOFError, OSError):
+            return
+
+        if eng.has_work():
+            eng.step()
+            finished = eng.collect_finished()
+            if token_buf:
+                conn.send(("tokens", token_buf))
+                token_buf = []
+            if finished:
+                now = time.monotonic()
+                out = []
+                for seq in finished:
+                    seq_by_id.pop(seq.request_id, None)
+                    stream_ids.discard(seq.request_id)
+                    out.append(
+                        FinishedRequest(
+                            request_id=seq.request_id,
+                            engine_name=engine_name,
+                            prompt_len=seq.prompt_len,
+                            output_token_ids=list(seq.output_token_ids),
+                            ttft=seq.ttft(),
+                            tpot=seq.tpot(),
+                            e2e=(seq.finish_time or now) - seq.arrival_time,
+                        )
+                    )
+                conn.send(("finished", out))
+            if mgr is not None:
+                resident = mgr.resident_names()
+                if resident != prev_resident:
+                    prev_resident = resident
+                    publish_adapters()
+        publish_gauges()
+
+
+class EngineHandle:
+    """Parent-side wrapper of one engine worker process."""
+
+    def __init__(
+        self,
+        pool: "EnginePool",
+        engine: Engine,
+        slot_base: int,
+        model_name: str,
+        device: str,
+        engine_kwargs: Dict[str, Any],
+        lora_rank: int,
+    ):
+        self.pool = pool
+        self.engine = engine
+        self.slot_base = slot_base
+        self.model_name = model_name
+        self.device = device
+        self.engine_kwargs = engine_kwargs
+        self.lora_rank = lora_rank
+        self.active_adapters: List[str] = []
+        self.max_slots = int(engine_kwargs.get("max_loras", 8))
+        self.healthy = False
+        self.proc: Optional[mp.process.BaseProcess] = None
+        self.conn = None
+        self._reader: Optional[threading.Thread] = None
+        # adapter registry for replay-on-restart: name -> (source, rank)
+        self.adapter_registry: Dict[str, Any] = {}
+        self._adapter_events: Dict[str, threading.Event] = {}
+        self._adapter_errors: Dict[str, Optional[str]] = {}
+
+    def start(self, ready_timeout_s: float = 300.0) -> None:
+        ctx = mp.get_context("spawn")
+        parent_conn, child_conn = ctx.Pipe()
+        self.conn = parent_conn
+        self.proc = ctx.Process(
+            target=_worker_main,
+            args=(
+                child_conn,
+                self.pool.shared,
+                self.slot_base,
+                self.engine.name,
+                self.model_name,
+                self.device,
+                self.engine_kwargs,
+                self.lora_rank,
+            ),
+            daemon=True,
+            name=f"engine-worker-{self.engine.name}",
+        )
+        self.proc.start()
+        child_conn.close()
+        if not parent_conn.poll(ready_timeout_s):
+            raise TimeoutError(f"engine {self.engine.name} did not become ready")
+        msg = parent_conn.recv()
+        assert msg[0] == "ready", msg
+        self.healthy = True
+        self._reader = threading.Thread(
+            target=self._read_loop, daemon=True, name=f"reader-{self.engine.name}"
+        )
+        self._reader.start()
+
+    def _read_loop(self) -> None:
+        while True:
+            try:
+                msg = self.conn.recv()
+            except (EOFError, OSError):
+                self.healthy = False
+                return
+            op = msg[0]
+            if op == "finished":
+                for fin in msg[1]:
+                    self.pool._on_finished(fin)
+            elif op == "tokens":
+                self.pool._on_tokens(msg[1])
+            elif op == "adapters":
+                self.active_adapters = msg[1]
+                self.max_slots = msg[2]
+            elif op == "adapter_loaded":
+                _, name, err = msg
+                self._adapter_errors[name] = err
+                ev = self._adapter_events.get(name)
+                if ev:
+                    ev.set()
+            elif op == "adapter_unloaded":
+                _, name, _ok = msg
+                ev = self._adapter_events.get(name)
+                if ev:
+                    ev.set()
+            elif op == "error":
+                self.pool._on_error(msg[1], msg[2])
+
+    # -- command API --------------------------------------------------------
+    def submit(self, req: PoolRequest) -> None:
+        self.conn.send(("add", req))
+
+    def abort(self, request_id: str) -> None:
+        self.conn.send(("abort", request_id))
+
+    def load_adapter(
+        self, name: str, source: Any = None, rank: int = 16, wait: bool = True,
+        timeout_s: float = 120.0,
+    ) -> Optional[str]:
+        """Hot-load one adapter (sidecar POST /v1/load_lora_adapter parity,
+        `sidecar.py:177-195`).  Returns an error string or None."""
+        self.adapter_registry[name] = (source, rank)
+        ev = threading.Event()
+        self._adapter_events[name] = ev
+        self.conn.send(("load_adapter", name, source, rank))
+        if not wait:
+            return None
+        if not ev.wait(timeout_s):
+            return f"timeout loading adapter {name}"
+        return self._adapter_errors.get(name)
+
+    def unload_adapter(self, name: str, wait: bool = True, timeout_s: float = 60.0) -> None:
+        """Sidecar POST /v1/unload_lora_adapter parity (`sidecar.py:197-213`)."""
+        self.adapter_registry.pop(name, None)
+        ev = threading.Event()
+        self._adapter_events[name] = ev
+        self.conn.send(("unload_adapter", name))
+        if wait:
+            ev.wait(timeout_s)
+
+    def drain(self, on: bool = True) -> None:
+        self.conn.send(("drain", on))
+
+    def stop(self) -> None:
+        try:
+            self.conn.send(("stop",))
+        except (BrokenPipeError, OSError):
+            pass
+        if self.proc is not None:
+            self.proc.join(timeout=10)
+            if self.proc.is_alive():
+                self.proc.terminate()
+        self.healthy = False
+
+    def kill(self) -> None:
+        if self.proc is not None:
+            self.proc.terminate()
+            self.proc.join(timeout=10)
+        self.healthy = False
+
+    # -- state reads --------------------------------------------------------
+    def metrics(self) -> EngineMetrics:
+        b = self.slot_base
+        s = self.pool.shared
+        return EngineMetrics(
+            active_models={n: 1 for n in self.active_adapters},
+            max_active_models=int(s[b + _SLOT_MAX_LORAS]),
+            running_queue_size=int(s[b + _SLOT_RUNNING]),
+            waiting_queue_size=int(s[b + _SLOT_WAITING]),
+            kv_cache_usage_percent=s[b + _SLOT_KV_USAGE],
+            kv_cache_max_token_capacity=int(s[b + _SLOT_KV_CAP]),
+        )
+
+    def heartbeat_age_s(self) -> float:
+        return time.time() - self.pool.shared[self.slot_base + _SLOT_HEARTBEAT]
+
+    def is_draining(self) -> bool:
+        return self.pool.shared[self.slot_base + _SLOT_DRAINING] > 0.5
+
+    def alive(self) -> bool:
+        return self.proc is not None and self.proc.is_alive() and self.healthy
+
+
+class EnginePool:
+    """N engine workers + shared gauges + routing entry point.
+
+    Satisfies the router's `EngineMetricsProvider` protocol directly from
+    shared memory — the 50 ms scrape staleness window of the reference
+    collapses to a memory read (SURVEY.md §5 'Distributed communication').
+    """
+
+    def __init__(
+        self,
+        model_name: str,
+        devices: Sequence[str],
+        engine_kwargs: Optional[Dict[str, Any]] = None,
+        lora_rank: int = 16,
+        datastore: Optional[Datastore] = None,
+        scheduler: Optional[Scheduler] = None,
+    ):
+        self.model_name = model_name
+        self.engine_kwargs = dict(engine_kwargs or {})
+        self.shared = mp.get_context("spawn").Array(
+            "d", SLOTS_PER_ENGINE * len(devices), lock=False
+        )
+        self.datastore = datastore or Datastore()
+        self.handles: Dict[str, EngineHandle] = {}
+        for i, dev in enumerate(devices):
+            eng = Engine(name=f"engine-{i}", gpu_id=i if dev.startswith("cuda") else -1)
+            self.handles[eng.name] = EngineHandle(
+                self,
+                eng,
+                slot_base=SLOTS_PER_ENGINE * i,
+                model_name=model_name,
+                device=dev,
+                engine_kwargs=self.engine_kwargs,
+                lora_rank=lora_rank,
+            )
+            self.datastore.add_engine(eng)
+        self.scheduler = scheduler or Scheduler(self)
+        self._tickets: Dict[str, RequestTicket] = {}
+        self._tickets_lock = threading.Lock()
+        self.on_finished: Optional[Callable[[FinishedRequest], None]] = None
+        self._watchdog: Optional[threading.Thread] = None
+        self._stop = threading.Event()
+
+    # -- lifecycle ----------------------------------------------------------
+    def start(self, watchdog: bool = True) -> None:
+        threads = []
+        for h in self.handles.values():
+            t = threading.Thread(target=h.start, daemon=True)
+            t.start()
+            threads.append(t)
+        for t in threads:
+            t.join()
+        for h in self.handles.values():
+            if not h.alive():
+                raise RuntimeError(f"engine {h.engine.name} failed to start")
+        if watchdog:
+            self._watchdog = threading.Thread(
+                target=self._watchdog_loop, daemon=True, name="pool-watchdog"
+            )
+            self._watchdog.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        for h in self.handles.values():
+            h.stop()
+
+    def _watchdog_loop(self, interval_s: float = 1.0) -> None:
+        """Engine health: stale heartbeat (hung kernel / dead worker) drops
+        the engine from the routing set — reference Ready-condition semantics
+        (`endpointslice_reconciler.go:107-111`)."""
+        while not self._stop.wait(interval_s):
+            for h in self.handles.values():
+                stale = h.heartbeat_age_s() > HEARTBEAT_STALE_S
+                dead = h.proc is not None and not h.proc.is_alive()
+                if (stale or dead) and h.healthy:
+                    log.error(
+                        "engine %s unhealthy (stale=%s dead=%s); dropping from pool",
+                        h.engine.name, stale, dead,
+                    )
+                    h.healthy = False
+                    self.datastore.remove_engine(h.engine.name)
+                    self._fail_inflight(h.engine.name, "engine died")
+
+    def _fail_inflight(self, engine_name: str, reason: str) -> None:
+        with self._tickets_lock:
+            doomed = [t for t in self._tickets.values() if t.engine_name == engine_name]
+            for t in doomed:
+                del self._tickets[t.request_id]
+        for t in doomed:
+            t.error = reason
+            if t.stream:
+                t.tokens.put(None)
+            t.done.set()
+
+    def restart_engine(self, name: str) -> None:
+        """Re-spawn a dead engine and replay its adapter registry (reference:
+        pod restart + sidecar reconcile, SURVEY.md §5 failure detection)."""
+        h = self.handles[name]
+        h.kill()
+        registry = dict(h.adapter_registry)
+        h.start()
+        for aname, (source, rank) in registry.items():
+            h.load_adapter(aname, source, rank, wait=False)
+        self.datastore.add_engine(h.engine)
+
+    # -- EngineMetricsProvider protocol -------------------------------------
+    def all_engine_metrics(self) -> List[EngineState]:
+        out = []
+        for e in self.datastore.get_engines():
+            h = self.handles.get(e.name)
+            if h is None or not h.healthy or h.is_draining():
+                continue
+            out.append(EngineState(engine=e, metrics=h.metrics()))
+        return out
+
+    # -- request path -------------------------------------------------------
+    def route(self, req: LLMRequest) -> Engine:
+        """Router decision only (reference Schedule, scheduler.go:113-122)."""
+        return self.scheduler.schedule(req)
+
+    def submit(self, engine_name: str, req: PoolRequest) -> RequestTicket:
+        h = self.handles[engine_name]
+        ticket = RequestTicket(req.request_id, engine_name, req.stream)
+        with self._tickets_lock:
+            self._tickets[req.request_id] = ticket
+        h.submit(req)
+        return ticket
+
+    def route_and_submit(self, lreq: LLMRequest, preq: PoolRequest) -> RequestTicket:
+        """The full hot path: filter-tree routing + direct dispatch (replaces
+        the reference's header-mutation + Envoy ORIGINAL_DST hop).  Raises
+        ResourceExhausted for shed sheddable traffic (HTTP 429 upstream)."""
+        engine = self.route(lreq)
+        return self.submit(engine.name, preq)
+
+    def abort(self, request_id: str) -> None:
+        with self._tickets_lock:
+            ticket = self._tickets.pop(request_id, None)
+        if ticket is not None:
+            self.handles[ticket.engine_name].abort(request_id)
+            ticket.error = "aborted"
+            if ticket.stream:
+                ticket.tokens.put(None)
+            ticket.done.set()
+
+    # -- worker event sinks --------------------------------------------------
+    def _on_finished(self, fin: FinishedRequest) -> None:
+        with self._tickets_lock:
+            ticket = self._tickets.pop(fin.request_id, None)
+        if ticket is not None:
+            ticket.result = fin
+            if ticket.stream:
+                ticket.tokens.put(None)  # sentinel: stream end
+            ticket.done.set()
+        if self.on_finished is not None:
+            self.on_finished(fin)
+
+    def _on_tokens(self, toks: List) -> None:
+        with self._tickets_lock:
+            for rid, tok in toks:
+                t = self._tickets.get(rid)
+                if t is not None:
+                    t.tokens.put(tok)
+
+    def _on_error(self, request_id: Optional[str], msg: str) -> None:
+        if request_id is None:
+            log.error("engine error: %s", msg)
+            return
+        with self._tickets_lock:
+            ticket = self._tickets.pop(request_id, None)
+        if ticket is not None:
+            ticket.error = msg
+            if ticket.stream:
+                ticket.tokens.put(None)
+            ticket.done.set()
