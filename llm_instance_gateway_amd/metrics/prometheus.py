@@ -25,6 +25,7 @@ Two parity roles:
 from __future__ import annotations
 
 import bisect
+import re
 import threading
 import time
 from typing import Dict, Iterable, List, Optional, Tuple
@@ -76,6 +77,10 @@ def render_engine_metrics(
     return "\n".join(lines) + "\n"
 
 
+_LABEL_RE = re.compile(r'([A-Za-z_][\w.:-]*)="((?:[^"\\]|\\.)*)"')
+_UNESCAPE_RE = re.compile(r"\\(.)")
+
+
 def _parse_line(line: str) -> Optional[Tuple[str, Dict[str, str], float]]:
     """Parse one Prometheus sample line -> (name, labels, value)."""
     line = line.strip()
@@ -85,33 +90,10 @@ def _parse_line(line: str) -> Optional[Tuple[str, Dict[str, str], float]]:
     if "{" in line:
         name, rest = line.split("{", 1)
         labpart, _, valpart = rest.rpartition("}")
-        # label parsing tolerant of commas inside quoted values
-        i, cur_key, cur_val, in_val, buf = 0, "", "", False, []
-        key = ""
-        while i < len(labpart):
-            c = labpart[i]
-            if not in_val:
-                if c == "=":
-                    key = "".join(buf).strip().strip(",")
-                    buf = []
-                    # expect opening quote
-                    i += 1
-                    if i < len(labpart) and labpart[i] == '"':
-                        in_val = True
-                    continue
-                buf.append(c)
-            else:
-                if c == "\\" and i + 1 < len(labpart):
-                    buf.append(labpart[i + 1])
-                    i += 2
-                    continue
-                if c == '"':
-                    labels[key] = "".join(buf)
-                    buf = []
-                    in_val = False
-                else:
-                    buf.append(c)
-            i += 1
+        for key, raw in _LABEL_RE.findall(labpart):
+            labels[key] = _UNESCAPE_RE.sub(
+                lambda m: {"n": "\n"}.get(m.group(1), m.group(1)), raw
+            )
         value_str = valpart.strip()
     else:
         parts = line.split()

@@ -358,11 +358,13 @@ class EngineHandle:
         self.conn.send(("abort", request_id))
 
     def load_adapter(
-        self, name: str, source: Any = None, rank: int = 16, wait: bool = True,
-        timeout_s: float = 120.0,
+        self, name: str, source: Any = None, rank: Optional[int] = None,
+        wait: bool = True, timeout_s: float = 120.0,
     ) -> Optional[str]:
         """Hot-load one adapter (sidecar POST /v1/load_lora_adapter parity,
         `sidecar.py:177-195`).  Returns an error string or None."""
+        if rank is None:
+            rank = self.lora_rank
         self.adapter_registry[name] = (source, rank)
         ev = threading.Event()
         self._adapter_events[name] = ev
