@@ -1,0 +1,114 @@
+"""Gateway server entrypoint.
+
+The reference ext-proc `main()` (`pkg/ext-proc/main.go:59-160`): flag parsing,
+datastore + reconcilers + provider + scheduler + gRPC server, graceful
+shutdown.  Here the same wiring, MI355X-native: build the per-GPU engine pool,
+attach the filter-tree router over shared-memory gauges, start the config
+reconciler (replaces the three K8s controllers), serve the OpenAI-compatible
+HTTP front door, drain on SIGTERM/SIGINT.
+
+Flags mirror `main.go:31-41` where they still make sense in-process
+(port, pool name, refresh intervals collapse to config), plus the
+config-file path that replaces the K8s API.
+
+    python -m llm_instance_gateway_amd.server --config gateway.yaml
+    python -m llm_instance_gateway_amd.server --model tiny --devices cpu,cpu
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import signal
+import sys
+from typing import List, Optional
+
+log = logging.getLogger(__name__)
+
+
+def build_pool(args, cfg) -> "EnginePool":
+    import torch
+
+    from .api.types import InferenceModel
+    from .router.filters import build_default_tree
+    from .router.scheduler import Scheduler
+    from .serving.pool import EnginePool
+
+    model = args.model or (cfg.pool.base_model if cfg else "") or "llama3-8b"
+    if args.devices:
+        devices: List[str] = args.devices.split(",")
+    elif cfg and cfg.pool.gpu_ids:
+        devices = [f"cuda:{g}" for g in cfg.pool.gpu_ids]
+    elif torch.cuda.is_available():
+        devices = [f"cuda:{i}" for i in range(torch.cuda.device_count())]
+    else:
+        devices = ["cpu"]
+
+    engine_kwargs = {}
+    if not any(d.startswith("cuda") for d in devices):
+        engine_kwargs = dict(
+            num_kv_blocks=256, max_num_seqs=16,
+            max_num_batched_tokens=512, max_model_len=512,
+        )
+    pool = EnginePool(model, devices=devices, engine_kwargs=engine_kwargs)
+    if cfg:
+        tree = build_default_tree(
+            kv_cache_threshold=cfg.kv_cache_threshold,
+            queue_threshold_critical=cfg.queue_threshold_critical,
+            queueing_threshold_lora=cfg.queueing_threshold_lora,
+        )
+        pool.scheduler = Scheduler(pool, tree=tree)
+    # the base model is always addressable (engines exist even with no config)
+    pool.datastore.store_model(InferenceModel(model_name=model))
+    return pool
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    ap = argparse.ArgumentParser(description=__doc__.splitlines()[0])
+    ap.add_argument("--config", default="", help="gateway YAML (pool/models/adapters/thresholds)")
+    ap.add_argument("--port", type=int, default=8081, help="HTTP front-door port (reference gateway.yaml:13-15)")
+    ap.add_argument("--model", default="", help="base model preset override")
+    ap.add_argument("--devices", default="", help="comma list, e.g. cuda:0,cuda:1 or cpu")
+    ap.add_argument("--v", type=int, default=1, help="log verbosity (klog analogue)")
+    args = ap.parse_args(argv)
+
+    logging.basicConfig(
+        level=logging.DEBUG if args.v >= 3 else logging.INFO,
+        format="%(asctime)s %(name)s %(levelname)s %(message)s",
+    )
+
+    from .api.config import ConfigReconciler, load_config
+    from .serving.frontdoor import build_app
+
+    cfg = load_config(args.config) if args.config else None
+    pool = build_pool(args, cfg)
+    log.info("starting %d engine(s)...", len(pool.handles))
+    pool.start()
+
+    reconciler = None
+    if args.config:
+        reconciler = ConfigReconciler(args.config, pool)
+        reconciler.start()
+
+    app = build_app(pool)
+
+    def shutdown(signum, frame):
+        log.info("signal %s: draining and stopping", signum)
+        if reconciler:
+            reconciler.stop()
+        pool.stop()
+        sys.exit(0)
+
+    signal.signal(signal.SIGTERM, shutdown)
+    signal.signal(signal.SIGINT, shutdown)
+
+    import uvicorn
+
+    port = args.port if args.port else (cfg.port if cfg else 8081)
+    uvicorn.run(app, host="0.0.0.0", port=port, log_level="warning")
+    pool.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

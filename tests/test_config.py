@@ -1,0 +1,193 @@
+"""Config plane tests.
+
+Mirrors the sidecar's reconcile tests (`tools/dynamic-lora-sidecar/sidecar/
+test_sidecar.py`: load ensureExist−ensureNotExist, skip both-lists dupes) and
+the reconciler unit tests (`backend/inferencemodel_reconciler_test.go`
+store/delete semantics), against a real CPU engine pool.
+"""
+
+import time
+
+import pytest
+import yaml
+
+from llm_instance_gateway_amd.api.config import (
+    AdapterSpec,
+    ConfigError,
+    ConfigReconciler,
+    GatewayConfig,
+    load_config,
+    parse_config,
+    reconcile_adapters,
+    reconcile_models,
+)
+from llm_instance_gateway_amd.api.types import Criticality
+from llm_instance_gateway_amd.router.datastore import Datastore
+from llm_instance_gateway_amd.serving import EnginePool
+
+CFG_YAML = """
+pool:
+  name: mi355x-pool
+  base_model: tiny
+  gpu_ids: [0, 1]
+models:
+  - model_name: chat
+    criticality: Critical
+    target_models:
+      - {name: chat, weight: 60}
+      - {name: lora-a, weight: 40}
+  - model_name: batch
+    criticality: Sheddable
+  - model_name: other-pool-model
+    criticality: Default
+    pool_ref: some-other-pool
+ensureExist:
+  models:
+    - id: lora-a
+      rank: 8
+    - id: lora-dupe
+ensureNotExist:
+  models:
+    - id: lora-gone
+    - id: lora-dupe
+thresholds:
+  kv_cache_threshold: 0.7
+  queue_threshold_critical: 3
+  queueing_threshold_lora: 25
+refresh:
+  metrics_interval_s: 0.01
+server:
+  port: 9099
+"""
+
+
+def test_parse_config_full():
+    cfg = parse_config(yaml.safe_load(CFG_YAML))
+    assert cfg.pool.name == "mi355x-pool"
+    assert cfg.pool.gpu_ids == [0, 1]
+    assert [m.model_name for m in cfg.models] == ["chat", "batch", "other-pool-model"]
+    assert cfg.models[0].criticality == Criticality.CRITICAL
+    assert cfg.models[0].target_models[1].weight == 40
+    assert [a.id for a in cfg.ensure_exist] == ["lora-a", "lora-dupe"]
+    assert cfg.ensure_exist[0].rank == 8
+    assert [a.id for a in cfg.ensure_not_exist] == ["lora-gone", "lora-dupe"]
+    assert cfg.kv_cache_threshold == 0.7
+    assert cfg.queue_threshold_critical == 3
+    assert cfg.queueing_threshold_lora == 25
+    assert cfg.refresh_metrics_interval_s == 0.01
+    assert cfg.port == 9099
+
+
+@pytest.mark.parametrize(
+    "raw",
+    [
+        {"pool": {"gpu_ids": [0]}},  # pool without name
+        {"models": [{"criticality": "Critical"}]},  # model without name
+        {"models": [{"model_name": "x", "criticality": "Bogus"}]},
+        {"thresholds": {"kv_cache_threshold": 1.5}},
+        {"ensureExist": {"models": [{"source": "p"}]}},  # adapter without id
+    ],
+)
+def test_parse_config_rejects(raw):
+    with pytest.raises(ConfigError):
+        parse_config(raw)
+
+
+def test_reconcile_models_store_and_delete():
+    """InferenceModelReconciler semantics: store models whose pool_ref matches,
+    delete vanished ones (`inferencemodel_reconciler.go:45-55`)."""
+    cfg = parse_config(yaml.safe_load(CFG_YAML))
+    ds = Datastore()
+    reconcile_models(cfg, ds)
+    assert ds.fetch_model("chat") is not None
+    assert ds.fetch_model("batch") is not None
+    # pool_ref mismatch filtered out (reference PoolRef gate)
+    assert ds.fetch_model("other-pool-model") is None
+    # now remove 'batch' from config -> deleted from datastore
+    cfg2 = GatewayConfig(pool=cfg.pool, models=[cfg.models[0]])
+    reconcile_models(cfg2, ds)
+    assert ds.fetch_model("batch") is None
+    assert ds.fetch_model("chat") is not None
+
+
+@pytest.fixture(scope="module")
+def pool():
+    p = EnginePool(
+        "tiny",
+        devices=["cpu"],
+        engine_kwargs=dict(
+            num_kv_blocks=64, max_num_seqs=4,
+            max_num_batched_tokens=128, max_model_len=128, max_loras=4,
+        ),
+        lora_rank=8,
+    )
+    p.start(watchdog=False)
+    yield p
+    p.stop()
+
+
+def test_reconcile_adapters_diff_and_dupe_skip(pool):
+    cfg = GatewayConfig(
+        ensure_exist=[AdapterSpec(id="lora-a", rank=8), AdapterSpec(id="lora-dupe", rank=8)],
+        ensure_not_exist=[AdapterSpec(id="lora-gone"), AdapterSpec(id="lora-dupe")],
+    )
+    loaded, unloaded, skipped = reconcile_adapters(cfg, pool)
+    assert loaded == {"lora-a"}
+    assert skipped == {"lora-dupe"}
+    h = pool.handles["engine-0"]
+    assert "lora-a" in h.adapter_registry
+    assert "lora-dupe" not in h.adapter_registry
+    # idempotent: second run loads nothing new
+    loaded2, _, _ = reconcile_adapters(cfg, pool)
+    assert loaded2 == set()
+    # flip lora-a to ensureNotExist -> unloaded
+    cfg2 = GatewayConfig(ensure_not_exist=[AdapterSpec(id="lora-a")])
+    _, unloaded2, _ = reconcile_adapters(cfg2, pool)
+    assert unloaded2 == {"lora-a"}
+    assert "lora-a" not in h.adapter_registry
+
+
+def test_config_reconciler_watches_file(tmp_path, pool):
+    path = tmp_path / "gw.yaml"
+    path.write_text(
+        "pool: {name: p, base_model: tiny}\n"
+        "models: [{model_name: watched-model}]\n"
+    )
+    rec = ConfigReconciler(
+        str(path), pool, poll_interval_s=0.05,
+        health_timeout_s=5, health_interval_s=0.05,
+    )
+    rec.start()
+    try:
+        assert rec.reconcile_count == 1
+        assert pool.datastore.fetch_model("watched-model") is not None
+        # mutate the file -> picked up by the polling observer
+        time.sleep(0.05)
+        path.write_text(
+            "pool: {name: p, base_model: tiny}\n"
+            "models: [{model_name: watched-model-v2}]\n"
+        )
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline:
+            if pool.datastore.fetch_model("watched-model-v2") is not None:
+                break
+            time.sleep(0.05)
+        assert pool.datastore.fetch_model("watched-model-v2") is not None
+        assert pool.datastore.fetch_model("watched-model") is None  # deleted
+        # malformed update is rejected; last good state stays
+        count = rec.reconcile_count
+        time.sleep(0.05)
+        path.write_text("models: [{criticality: Critical}]\n")
+        deadline = time.monotonic() + 3
+        while time.monotonic() < deadline and rec.reconcile_count == count:
+            time.sleep(0.05)
+        assert pool.datastore.fetch_model("watched-model-v2") is not None
+    finally:
+        rec.stop()
+
+
+def test_load_config_file(tmp_path):
+    p = tmp_path / "c.yaml"
+    p.write_text(CFG_YAML)
+    cfg = load_config(str(p))
+    assert cfg.pool.base_model == "tiny"
