@@ -99,8 +99,10 @@ class ModelRunner:
         )
         self.adapter_manager = None
         self.graph_runner = None
+        # MoE dispatch is token-dependent; a captured decode graph would
+        # freeze the capture batch's expert routing -> eager for MoE.
         self._use_graphs = (
-            self.device.type == "cuda" and not config.enforce_eager
+            self.device.type == "cuda" and not config.enforce_eager and not mc.is_moe
         )
 
     # -- batch construction ---------------------------------------------------
