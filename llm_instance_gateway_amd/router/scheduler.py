@@ -43,8 +43,16 @@ class Scheduler:
         self._rng = rng or random.Random()
 
     def schedule(self, req: LLMRequest) -> Engine:
-        """Reference Schedule (scheduler.go:113-122)."""
+        """Reference Schedule (scheduler.go:113-122).
+
+        Extension: if the request names a pool (mixed-pool serving — several
+        sub-pools behind one front door, BASELINE config #4), only that
+        pool's engines are candidates; untagged engines always qualify."""
         states = self.provider.all_engine_metrics()
+        if req.pool_ref:
+            states = [
+                s for s in states if s.engine.pool in ("", req.pool_ref)
+            ] or states
         try:
             survivors = self.filter.apply(req, states)
         except DropRequest as e:

@@ -28,6 +28,13 @@ class Engine:
     name: str
     gpu_id: int = -1
     address: str = ""
+    # Extensions over the reference (mixed-pool serving, BASELINE config #4):
+    # the base model this engine serves and the pool it belongs to.  The
+    # reference needs neither because one InferencePool == one homogeneous
+    # fleet; here several sub-pools (8B TP=1 engines + 70B TP=4 engine) can
+    # coexist behind one front door.
+    model: str = ""
+    pool: str = ""
 
 
 @dataclass
@@ -68,8 +75,9 @@ class LLMRequest:
     target_models: Dict[str, int] = field(default_factory=dict)
     resolved_target_model: str = ""
     critical: bool = False
-    # Extensions over the reference (used by admission/SLO accounting, not by
-    # the parity filter chain):
+    # Extensions over the reference (used by admission/SLO accounting and
+    # mixed-pool candidate selection, not by the parity filter chain):
     prompt_len: int = 0
     max_new_tokens: int = 0
     request_id: str = ""
+    pool_ref: str = ""

@@ -1,4 +1,4 @@
 from .frontdoor import build_app
-from .pool import EnginePool, FinishedRequest, PoolRequest, RequestTicket
+from .pool import EnginePool, EngineSpec, FinishedRequest, PoolRequest, RequestTicket
 
-__all__ = ["EnginePool", "FinishedRequest", "PoolRequest", "RequestTicket", "build_app"]
+__all__ = ["EnginePool", "EngineSpec", "FinishedRequest", "PoolRequest", "RequestTicket", "build_app"]

@@ -129,6 +129,7 @@ class Gateway:
             critical=critical,
             prompt_len=len(prompt_ids),
             max_new_tokens=req.max_tokens,
+            pool_ref=im.pool_ref,
         )
         preq = PoolRequest(
             request_id=req.request_id or f"cmpl-{uuid.uuid4().hex[:24]}",

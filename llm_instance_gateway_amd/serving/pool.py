@@ -83,6 +83,19 @@ class FinishedRequest:
     finish_reason: str = "length"
 
 
+@dataclass
+class EngineSpec:
+    """One engine of a (possibly mixed) pool: base model, device set
+    (len == TP degree), per-engine knobs."""
+
+    name: str
+    model: str
+    devices: List[str]
+    engine_kwargs: Dict[str, Any] = field(default_factory=dict)
+    lora_rank: int = 16
+    pool_ref: str = ""
+
+
 class RequestTicket:
     """Parent-side handle for one in-flight request (thread-safe; the async
     front door bridges it onto the event loop)."""
@@ -114,8 +127,17 @@ def _worker_main(
     device: str,
     engine_kwargs: Dict[str, Any],
     lora_rank: int,
+    tp_rank: int = 0,
+    tp_size: int = 1,
+    master_port: int = 0,
 ) -> None:
-    """Engine worker process body: one LLMEngine, one GPU, one event loop."""
+    """Engine worker process body: one LLMEngine, one GPU, one event loop.
+
+    TP engines (tp_size > 1, BASELINE config #4's 70B TP=4) run this body on
+    every rank with its own process group (gloo on CPU, RCCL over xGMI on
+    GPU): rank 0 fronts the request pipe and broadcasts each iteration's
+    command batch; all ranks then run the same deterministic scheduler +
+    model step, synchronised by the per-layer collectives."""
     import torch
 
     from ..engine import EngineConfig, LLMEngine, SamplingParams
@@ -124,10 +146,22 @@ def _worker_main(
 
     if device.startswith("cuda"):
         torch.cuda.set_device(device)
+    tp = None
+    if tp_size > 1:
+        from ..parallel import TPGroup, init_distributed
+
+        init_distributed(
+            backend="nccl" if device.startswith("cuda") else "gloo",
+            rank=tp_rank,
+            world_size=tp_size,
+            master_port=master_port,
+        )
+        tp = TPGroup(device=torch.device(device))
+    is_front = tp_rank == 0
     mc = get_model_config(model_name)
     cfg = EngineConfig(model=mc, engine_id=engine_name, **engine_kwargs)
     torch.manual_seed(1234)  # same base weights on every engine of the pool
-    eng = LLMEngine(cfg, device=device)
+    eng = LLMEngine(cfg, device=device, tp=tp)
     mgr: Optional[AdapterManager] = None
     if cfg.max_loras > 0:
         mgr = AdapterManager(eng, max_slots=cfg.max_loras, rank=lora_rank)
@@ -149,9 +183,13 @@ def _worker_main(
         shared[b + _SLOT_MAX_LORAS] = float(m.max_active_models)
         shared[b + _SLOT_DRAINING] = 1.0 if draining else 0.0
 
+    def send(msg) -> None:
+        if is_front:
+            conn.send(msg)
+
     def publish_adapters() -> None:
         if mgr is not None:
-            conn.send(("adapters", mgr.resident_names(), mgr.max_slots))
+            send(("adapters", mgr.resident_names(), mgr.max_slots))
 
     def on_token(seq, tok):
         if seq.request_id in stream_ids:
@@ -159,80 +197,106 @@ def _worker_main(
 
     eng.on_token = on_token
 
-    publish_gauges()
-    conn.send(("ready",))
+    # TP: commands are broadcast from the front rank over a dedicated gloo
+    # group (host-side metadata must not ride the RCCL compute stream).
+    tp_cmd = None
+    if tp is not None:
+        import torch.distributed as dist
+
+        from ..parallel import TPGroup
+
+        tp_cmd = TPGroup(group=dist.new_group(backend="gloo"))
+
+    nonlocal_draining = [draining]
+
+    def apply_cmd(msg) -> bool:
+        """Apply one command on this rank; False means stop."""
+        op = msg[0]
+        if op == "add":
+            r: PoolRequest = msg[1]
+            if r.stream and is_front:
+                stream_ids.add(r.request_id)
+            try:
+                seq = eng.add_request(
+                    r.prompt_token_ids,
+                    SamplingParams(
+                        max_new_tokens=r.max_new_tokens,
+                        temperature=r.temperature,
+                        top_p=r.top_p,
+                    ),
+                    lora_name=r.lora_name,
+                    request_id=r.request_id,
+                    critical=r.critical,
+                    arrival_time=r.arrival_time,
+                )
+                seq_by_id[r.request_id] = seq
+            except Exception as e:  # e.g. prompt > max_model_len
+                send(("error", r.request_id, str(e)))
+        elif op == "abort":
+            eng.abort_request(msg[1])
+        elif op == "load_adapter":
+            _, name, source, rank = msg
+            try:
+                if mgr is None:
+                    raise RuntimeError("engine has no adapter slots")
+                if isinstance(source, dict):
+                    mgr.register_weights(name, source, rank=rank)
+                elif isinstance(source, str) and os.path.exists(source):
+                    weights = torch.load(source, weights_only=True)
+                    mgr.register_weights(name, weights, rank=rank)
+                else:
+                    mgr.register_random(
+                        name, seed=abs(hash(name)) % (2**31), rank=rank
+                    )
+                mgr.prepare([name])
+                send(("adapter_loaded", name, None))
+            except Exception as e:
+                send(("adapter_loaded", name, str(e)))
+            publish_adapters()
+        elif op == "unload_adapter":
+            name = msg[1]
+            ok = mgr.unregister(name) if mgr is not None else False
+            send(("adapter_unloaded", name, ok))
+            publish_adapters()
+        elif op == "drain":
+            nonlocal_draining[0] = bool(msg[1])
+        elif op == "ping":
+            send(("pong", msg[1]))
+        elif op == "stop":
+            return False
+        return True
+
+    if is_front:
+        publish_gauges()
+        conn.send(("ready",))
     prev_resident: List[str] = []
 
     while True:
-        # drain all pending commands, then (maybe) step
-        timeout = 0.0 if eng.has_work() else 0.002
-        try:
-            while conn.poll(timeout):
-                msg = conn.recv()
-                op = msg[0]
-                if op == "add":
-                    r: PoolRequest = msg[1]
-                    if r.stream:
-                        stream_ids.add(r.request_id)
-                    try:
-                        seq = eng.add_request(
-                            r.prompt_token_ids,
-                            SamplingParams(
-                                max_new_tokens=r.max_new_tokens,
-                                temperature=r.temperature,
-                                top_p=r.top_p,
-                            ),
-                            lora_name=r.lora_name,
-                            request_id=r.request_id,
-                            critical=r.critical,
-                            arrival_time=r.arrival_time,
-                        )
-                        seq_by_id[r.request_id] = seq
-                    except Exception as e:  # e.g. prompt > max_model_len
-                        conn.send(("error", r.request_id, str(e)))
-                elif op == "abort":
-                    eng.abort_request(msg[1])
-                elif op == "load_adapter":
-                    _, name, source, rank = msg
-                    try:
-                        if mgr is None:
-                            raise RuntimeError("engine has no adapter slots")
-                        if isinstance(source, dict):
-                            mgr.register_weights(name, source, rank=rank)
-                        elif isinstance(source, str) and os.path.exists(source):
-                            weights = torch.load(source, weights_only=True)
-                            mgr.register_weights(name, weights, rank=rank)
-                        else:
-                            mgr.register_random(
-                                name, seed=abs(hash(name)) % (2**31), rank=rank
-                            )
-                        mgr.prepare([name])
-                        conn.send(("adapter_loaded", name, None))
-                    except Exception as e:
-                        conn.send(("adapter_loaded", name, str(e)))
-                    publish_adapters()
-                elif op == "unload_adapter":
-                    name = msg[1]
-                    ok = mgr.unregister(name) if mgr is not None else False
-                    conn.send(("adapter_unloaded", name, ok))
-                    publish_adapters()
-                elif op == "drain":
-                    draining = bool(msg[1])
-                elif op == "ping":
-                    conn.send(("pong", msg[1]))
-                elif op == "stop":
-                    return
-                timeout = 0.0
-        except (EOFError, OSError):
-            return
+        # front rank drains its pipe into a command batch ...
+        cmds: List = []
+        if is_front:
+            timeout = 0.0 if eng.has_work() else 0.002
+            try:
+                while conn.poll(timeout):
+                    cmds.append(conn.recv())
+                    timeout = 0.0
+            except (EOFError, OSError):
+                cmds.append(("stop",))
+        # ... and every TP rank applies the identical batch in order
+        if tp_cmd is not None:
+            cmds = tp_cmd.broadcast_obj(cmds)
+        for msg in cmds:
+            if not apply_cmd(msg):
+                return
+        draining = nonlocal_draining[0]
 
         if eng.has_work():
             eng.step()
             finished = eng.collect_finished()
-            if token_buf:
+            if is_front and token_buf:
                 conn.send(("tokens", token_buf))
                 token_buf = []
-            if finished:
+            if finished and is_front:
                 now = time.monotonic()
                 out = []
                 for seq in finished:
@@ -250,12 +314,13 @@ def _worker_main(
                         )
                     )
                 conn.send(("finished", out))
-            if mgr is not None:
+            if mgr is not None and is_front:
                 resident = mgr.resident_names()
                 if resident != prev_resident:
                     prev_resident = resident
                     publish_adapters()
-        publish_gauges()
+        if is_front:
+            publish_gauges()
 
 
 class EngineHandle:
@@ -267,7 +332,7 @@ class EngineHandle:
         engine: Engine,
         slot_base: int,
         model_name: str,
-        device: str,
+        devices: List[str],
         engine_kwargs: Dict[str, Any],
         lora_rank: int,
     ):
@@ -275,13 +340,15 @@ class EngineHandle:
         self.engine = engine
         self.slot_base = slot_base
         self.model_name = model_name
-        self.device = device
+        # len(devices) == tensor-parallel degree; devices[0] hosts the front
+        # rank that owns the request pipe (BASELINE config #4's TP=4 engine)
+        self.devices = list(devices)
         self.engine_kwargs = engine_kwargs
         self.lora_rank = lora_rank
         self.active_adapters: List[str] = []
         self.max_slots = int(engine_kwargs.get("max_loras", 8))
         self.healthy = False
-        self.proc: Optional[mp.process.BaseProcess] = None
+        self.procs: List[mp.process.BaseProcess] = []
         self.conn = None
         self._reader: Optional[threading.Thread] = None
         # adapter registry for replay-on-restart: name -> (source, rank)
@@ -289,26 +356,48 @@ class EngineHandle:
         self._adapter_events: Dict[str, threading.Event] = {}
         self._adapter_errors: Dict[str, Optional[str]] = {}
 
+    @property
+    def tp_size(self) -> int:
+        return len(self.devices)
+
+    @property
+    def proc(self) -> Optional[mp.process.BaseProcess]:
+        return self.procs[0] if self.procs else None
+
     def start(self, ready_timeout_s: float = 300.0) -> None:
         ctx = mp.get_context("spawn")
         parent_conn, child_conn = ctx.Pipe()
         self.conn = parent_conn
-        self.proc = ctx.Process(
-            target=_worker_main,
-            args=(
-                child_conn,
-                self.pool.shared,
-                self.slot_base,
-                self.engine.name,
-                self.model_name,
-                self.device,
-                self.engine_kwargs,
-                self.lora_rank,
-            ),
-            daemon=True,
-            name=f"engine-worker-{self.engine.name}",
-        )
-        self.proc.start()
+        master_port = 0
+        if self.tp_size > 1:
+            # each TP engine is its own process-group world on 127.0.0.1
+            import socket
+
+            with socket.socket() as s:
+                s.bind(("127.0.0.1", 0))
+                master_port = s.getsockname()[1]
+        self.procs = []
+        for r, dev in enumerate(self.devices):
+            p = ctx.Process(
+                target=_worker_main,
+                args=(
+                    child_conn if r == 0 else None,
+                    self.pool.shared,
+                    self.slot_base,
+                    self.engine.name,
+                    self.model_name,
+                    dev,
+                    self.engine_kwargs,
+                    self.lora_rank,
+                    r,
+                    self.tp_size,
+                    master_port,
+                ),
+                daemon=True,
+                name=f"engine-worker-{self.engine.name}-tp{r}",
+            )
+            p.start()
+            self.procs.append(p)
         child_conn.close()
         if not parent_conn.poll(ready_timeout_s):
             raise TimeoutError(f"engine {self.engine.name} did not become ready")
@@ -392,16 +481,16 @@ class EngineHandle:
             self.conn.send(("stop",))
         except (BrokenPipeError, OSError):
             pass
-        if self.proc is not None:
-            self.proc.join(timeout=10)
-            if self.proc.is_alive():
-                self.proc.terminate()
+        for p in self.procs:
+            p.join(timeout=10)
+            if p.is_alive():
+                p.terminate()
         self.healthy = False
 
     def kill(self) -> None:
-        if self.proc is not None:
-            self.proc.terminate()
-            self.proc.join(timeout=10)
+        for p in self.procs:
+            p.terminate()
+            p.join(timeout=10)
         self.healthy = False
 
     # -- state reads --------------------------------------------------------
@@ -424,7 +513,11 @@ class EngineHandle:
         return self.pool.shared[self.slot_base + _SLOT_DRAINING] > 0.5
 
     def alive(self) -> bool:
-        return self.proc is not None and self.proc.is_alive() and self.healthy
+        return (
+            bool(self.procs)
+            and all(p.is_alive() for p in self.procs)
+            and self.healthy
+        )
 
 
 class EnginePool:
@@ -437,30 +530,52 @@ class EnginePool:
 
     def __init__(
         self,
-        model_name: str,
-        devices: Sequence[str],
+        model_name: str = "",
+        devices: Sequence[str] = (),
         engine_kwargs: Optional[Dict[str, Any]] = None,
         lora_rank: int = 16,
         datastore: Optional[Datastore] = None,
         scheduler: Optional[Scheduler] = None,
+        specs: Optional[Sequence["EngineSpec"]] = None,
     ):
-        self.model_name = model_name
+        """Homogeneous pool: ``EnginePool(model, devices=[...])`` — one TP=1
+        engine per device (the reference's replica fleet).  Mixed pool:
+        ``EnginePool(specs=[EngineSpec(...), ...])`` — arbitrary per-engine
+        model / device-set / TP degree (BASELINE config #4)."""
+        if specs is None:
+            specs = [
+                EngineSpec(
+                    name=f"engine-{i}",
+                    model=model_name,
+                    devices=[dev],
+                    engine_kwargs=dict(engine_kwargs or {}),
+                    lora_rank=lora_rank,
+                )
+                for i, dev in enumerate(devices)
+            ]
+        self.model_name = model_name or (specs[0].model if specs else "")
         self.engine_kwargs = dict(engine_kwargs or {})
         self.shared = mp.get_context("spawn").Array(
-            "d", SLOTS_PER_ENGINE * len(devices), lock=False
+            "d", SLOTS_PER_ENGINE * len(specs), lock=False
         )
         self.datastore = datastore or Datastore()
         self.handles: Dict[str, EngineHandle] = {}
-        for i, dev in enumerate(devices):
-            eng = Engine(name=f"engine-{i}", gpu_id=i if dev.startswith("cuda") else -1)
+        for i, spec in enumerate(specs):
+            dev0 = spec.devices[0]
+            eng = Engine(
+                name=spec.name,
+                gpu_id=i if dev0.startswith("cuda") else -1,
+                model=spec.model,
+                pool=spec.pool_ref,
+            )
             self.handles[eng.name] = EngineHandle(
                 self,
                 eng,
                 slot_base=SLOTS_PER_ENGINE * i,
-                model_name=model_name,
-                device=dev,
-                engine_kwargs=self.engine_kwargs,
-                lora_rank=lora_rank,
+                model_name=spec.model,
+                devices=spec.devices,
+                engine_kwargs=dict(spec.engine_kwargs),
+                lora_rank=spec.lora_rank,
             )
             self.datastore.add_engine(eng)
         self.scheduler = scheduler or Scheduler(self)
@@ -500,7 +615,7 @@ class EnginePool:
         while not self._stop.wait(interval_s):
             for h in self.handles.values():
                 stale = h.heartbeat_age_s() > HEARTBEAT_STALE_S
-                dead = h.proc is not None and not h.proc.is_alive()
+                dead = bool(h.procs) and any(not p.is_alive() for p in h.procs)
                 if (stale or dead) and h.healthy:
                     log.error(
                         "engine %s unhealthy (stale=%s dead=%s); dropping from pool",
