@@ -63,8 +63,22 @@ class EngineState:
     metrics: EngineMetrics = field(default_factory=EngineMetrics)
 
     def clone(self) -> "EngineState":
-        """Deep copy (reference `PodMetrics.Clone`, types.go:39-53)."""
-        return EngineState(engine=self.engine, metrics=copy.deepcopy(self.metrics))
+        """Deep copy (reference `PodMetrics.Clone`, types.go:39-53).  Explicit
+        field copy, not copy.deepcopy: clone runs per engine per schedule()
+        call on the router hot path, and deepcopy's generic reflection is
+        ~30x slower than rebuilding the one dict + five scalars."""
+        m = self.metrics
+        return EngineState(
+            engine=self.engine,
+            metrics=EngineMetrics(
+                active_models=dict(m.active_models),
+                max_active_models=m.max_active_models,
+                running_queue_size=m.running_queue_size,
+                waiting_queue_size=m.waiting_queue_size,
+                kv_cache_usage_percent=m.kv_cache_usage_percent,
+                kv_cache_max_token_capacity=m.kv_cache_max_token_capacity,
+            ),
+        )
 
 
 @dataclass

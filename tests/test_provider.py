@@ -69,10 +69,18 @@ def test_direct_mode_update():
     assert p.all_engine_metrics()[0].metrics.waiting_queue_size == 9
 
 
-def test_snapshot_is_a_clone():
+def test_snapshot_isolated_from_updates():
+    """Snapshots are shared read-only references; writers REPLACE the state
+    object (atomic swap, provider.py all_engine_metrics), so a snapshot taken
+    before an update is unaffected by it."""
     ds = Datastore(engines=[Engine("e0", 0)])
     p = Provider(None, ds)
     p.init(start_loops=False)
     snap = p.all_engine_metrics()[0]
-    snap.metrics.waiting_queue_size = 123
-    assert p.all_engine_metrics()[0].metrics.waiting_queue_size == 0
+    p.update_engine_metrics(Engine("e0", 0), EngineMetrics(waiting_queue_size=9))
+    assert snap.metrics.waiting_queue_size == 0  # old snapshot untouched
+    assert p.all_engine_metrics()[0].metrics.waiting_queue_size == 9
+    # explicit clone() still available for callers that must mutate
+    c = snap.clone()
+    c.metrics.waiting_queue_size = 123
+    assert snap.metrics.waiting_queue_size == 0
