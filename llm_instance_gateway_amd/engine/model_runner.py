@@ -17,6 +17,7 @@ import torch
 from ..models.config import ModelConfig
 from .config import EngineConfig
 from .metadata import BatchMeta
+from ..utils.tracing import gpu_range
 from .sampling import sample_tokens
 from .scheduler import ScheduledSeq, SchedulerOutput
 from .sequence import Sequence
@@ -242,16 +243,20 @@ class ModelRunner:
         if out.is_empty:
             return {}
         if self._use_graphs and all(ss.num_tokens == 1 for ss in out.scheduled):
-            res = self._execute_graph(out)
+            with gpu_range("graph_replay"):
+                res = self._execute_graph(out)
             if res is not None:
                 return res
-        ids, pos, meta, sample_seqs, sample_idx, ordered = self._build_batch(out)
-        hidden = self.model(ids, pos, self.kv_caches, meta)
+        with gpu_range("build_batch"):
+            ids, pos, meta, sample_seqs, sample_idx, ordered = self._build_batch(out)
+        with gpu_range("forward"):
+            hidden = self.model(ids, pos, self.kv_caches, meta)
         sampled: Dict[int, int] = {}
         if sample_seqs:
-            idx = torch.tensor(sample_idx, dtype=torch.int64, device=self.device)
-            logits = self.model.compute_logits(hidden, idx)
-            tokens = sample_tokens(logits, sample_seqs)
+            with gpu_range("sample"):
+                idx = torch.tensor(sample_idx, dtype=torch.int64, device=self.device)
+                logits = self.model.compute_logits(hidden, idx)
+                tokens = sample_tokens(logits, sample_seqs)
             for seq, tok in zip(sample_seqs, tokens):
                 sampled[seq.seq_id] = tok
         for ss in ordered:

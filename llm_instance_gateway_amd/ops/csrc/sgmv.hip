@@ -9,48 +9,51 @@
 //   shrink:  u[t, :R]  = A[slot_t] (R x H)  · x[t]      (R = lora rank, <=64)
 //   expand:  y[t, :O] += s * B[slot_t] (O x R) · u[t]
 //
-// Both are memory-bound on the adapter weights; tokens sharing an adapter hit
-// the same A/B rows in L2 (256 MiB LLC holds tens of rank-16 adapter pairs).
+// Adapter working sets are small (32 rank-16 adapters' A for one module =
+// 4 MiB) so both kernels run out of L2 — the binding constraint is LATENCY,
+// not bandwidth.  v1 (one block per token / one y-vector per thread with
+// serialized loads) sat at ~6% of effective bandwidth: with T=128 decode
+// batches it put only 512 waves on a 256-CU chip with ~2 loads in flight
+// each.  v2 maximizes occupancy x memory-level parallelism:
 //
-// shrink: one workgroup per token; wave w handles rank rows w, w+4, ... —
-//   each lane covers H/64 dims (vectorized u16x8), wave-reduce per row.
-// expand: grid-stride over (token, 16-wide output chunks); u[t] (R values)
-//   is read via L1; each thread accumulates 16 outputs from B rows.
-
+// shrink: one WAVE per (token, rank-row) -> T*R waves (2048 at T=128);
+//   lanes stride the H dim in u16x8 vectors, 4x unrolled so 8 independent
+//   16 B loads are in flight per wave; wave-reduce, lane 0 writes.
+// expand: one thread per 8 consecutive outputs of one token; ALL B-row
+//   vectors (8 * R/8), u[t] and y are loaded before any FMA so ~20
+//   independent loads are outstanding per thread.
 #include "common.h"
 
 __global__ void sgmv_shrink_kernel(
-    uint16_t* __restrict__ u,        // [T, R] bf16 (zero-filled by caller)
+    uint16_t* __restrict__ u,        // [T, R] bf16
     const uint16_t* __restrict__ x,  // [T, H]
     const uint16_t* __restrict__ a,  // [S, R, H]
     const int* __restrict__ slots,   // [T]
     const int T, const int R, const int H) {
-  const int t = blockIdx.x;
+  // global wave id -> (token, rank row)
+  const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int t = wid / R;
+  if (t >= T) return;
+  const int r = wid - t * R;
   const int s = slots[t];
   if (s < 0) return;
-  const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int nwaves = blockDim.x >> 6;
   const uint16_t* xrow = x + (size_t)t * H;
+  const uint16_t* arow = a + ((size_t)s * R + r) * H;
   const int chunks = H / 8;  // 8 bf16 per lane-step
 
-  for (int r = wave; r < R; r += nwaves) {
-    const uint16_t* arow = a + ((size_t)s * R + r) * H;
-    float acc = 0.0f;
-    for (int c = lane; c < chunks; c += 64) {
-      f32x8 xv = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(xrow + c * 8));
-      f32x8 av = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(arow + c * 8));
+  float acc = 0.0f;
+#pragma unroll 4
+  for (int c = lane; c < chunks; c += 64) {
+    f32x8 xv = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(xrow + c * 8));
+    f32x8 av = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(arow + c * 8));
 #pragma unroll
-      for (int e = 0; e < 8; ++e) acc += xv[e] * av[e];
-    }
-    acc = wave_reduce_sum(acc);
-    if (lane == 0) u[(size_t)t * R + r] = f32_to_bf16(acc);
+    for (int e = 0; e < 8; ++e) acc += xv[e] * av[e];
   }
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) u[(size_t)t * R + r] = f32_to_bf16(acc);
 }
 
-// R is a template parameter (multiple of 8) so every B-row and u read is a
-// full 16 B u16x8 vector load — the scalar-2B-load version of this kernel
-// was 73% of total step time (guide common-mistake #2: never scalar bf16).
 template <int R>
 __global__ void sgmv_expand_kernel(
     uint16_t* __restrict__ y,        // [T, O] accumulated in place
@@ -67,18 +70,30 @@ __global__ void sgmv_expand_kernel(
     const long t = idx / (O / 8);
     const int s = slots[t];
     if (s < 0) continue;
-    // u[t]: R/8 vector loads; L1-hot across the token's O/8 threads
+    const uint16_t* brow = b + ((size_t)s * O + oc * 8) * R;
+    // issue EVERY load before any arithmetic: 8*(R/8) B vectors + u + y
+    u16x8 breg[8 * (R / 8)];
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+#pragma unroll
+      for (int rc = 0; rc < R / 8; ++rc)
+        breg[e * (R / 8) + rc] = *reinterpret_cast<const u16x8*>(
+            brow + (size_t)e * R + rc * 8);
+    u16x8 ureg[R / 8];
+#pragma unroll
+    for (int rc = 0; rc < R / 8; ++rc)
+      ureg[rc] =
+          *reinterpret_cast<const u16x8*>(u + (size_t)t * R + rc * 8);
+    u16x8 yv = *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8);
+
     float uv[R];
 #pragma unroll
     for (int rc = 0; rc < R / 8; ++rc) {
-      f32x8 v = bf16x8_to_f32x8(
-          *reinterpret_cast<const u16x8*>(u + (size_t)t * R + rc * 8));
+      f32x8 v = bf16x8_to_f32x8(ureg[rc]);
 #pragma unroll
       for (int e = 0; e < 8; ++e) uv[rc * 8 + e] = v[e];
     }
-    const uint16_t* brow = b + ((size_t)s * O + oc * 8) * R;
     f32x8 acc;
-    u16x8 yv = *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8);
 #pragma unroll
     for (int e = 0; e < 8; ++e) acc[e] = bf16_to_f32(yv[e]);
 #pragma unroll
@@ -86,8 +101,7 @@ __global__ void sgmv_expand_kernel(
       float a = 0.0f;
 #pragma unroll
       for (int rc = 0; rc < R / 8; ++rc) {
-        f32x8 bv = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(
-            brow + (size_t)e * R + rc * 8));
+        f32x8 bv = bf16x8_to_f32x8(breg[e * (R / 8) + rc]);
 #pragma unroll
         for (int q = 0; q < 8; ++q) a += bv[q] * uv[rc * 8 + q];
       }
@@ -102,10 +116,13 @@ extern "C" {
 void lig_sgmv_shrink_launch(uint16_t* u, const uint16_t* x, const uint16_t* a,
                             const int* slots, int T, int R, int H,
                             hipStream_t stream) {
-  dim3 grid(T);
+  // 4 waves per block, one wave per (token, rank-row)
+  const long waves = (long)T * R;
+  long grid = (waves + 3) / 4;
+  if (grid > 1048576) grid = 1048576;
   dim3 block(256);
-  hipLaunchKernelGGL(sgmv_shrink_kernel, grid, block, 0, stream, u, x, a,
-                     slots, T, R, H);
+  hipLaunchKernelGGL(sgmv_shrink_kernel, dim3((int)grid), block, 0, stream, u,
+                     x, a, slots, T, R, H);
 }
 
 void lig_sgmv_expand_launch(uint16_t* y, const uint16_t* u, const uint16_t* b,

@@ -39,6 +39,7 @@ from ..metrics.prometheus import GatewayMetrics, render_engine_metrics
 from ..router.scheduler import ResourceExhausted, SchedulingError
 from ..router.types import LLMRequest
 from ..utils import tokenizer
+from ..utils.tracing import SpanLog
 from .pool import EnginePool, PoolRequest, RequestTicket
 
 # Debug header the reference sets on every response (handlers/response.go:13-38).
@@ -101,6 +102,8 @@ class Gateway:
     def __init__(self, pool: EnginePool):
         self.pool = pool
         self.metrics = GatewayMetrics()
+        # per-request lifecycle spans (reference has none — SURVEY.md §5)
+        self.spans = SpanLog()
 
     # -- request resolution (reference HandleRequestBody, request.go:19-70) --
     def resolve(self, model_name: str) -> tuple:
@@ -141,8 +144,11 @@ class Gateway:
             critical=critical,
             stream=req.stream,
         )
+        self.spans.event(preq.request_id, "arrived", model=req.model,
+                         target=target, critical=critical)
         ticket = self.pool.route_and_submit(lreq, preq)
         self.metrics.route_latency.observe(time.monotonic() - t0)
+        self.spans.event(preq.request_id, "routed", engine=ticket.engine_name)
         return ticket, target
 
     def record_finished(self, ticket: RequestTicket, model: str) -> None:
@@ -157,6 +163,9 @@ class Gateway:
         if fin.tpot is not None:
             self.metrics.tpot.observe(fin.tpot)
         self.metrics.e2e.observe(fin.e2e)
+        self.spans.event(fin.request_id, "finished", engine=fin.engine_name,
+                         ttft=fin.ttft, e2e=fin.e2e,
+                         output_tokens=len(fin.output_token_ids))
 
 
 def _usage(prompt_len: int, completion_len: int) -> Dict[str, int]:
@@ -408,6 +417,11 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
             }
             for h in pool.handles.values()
         ]
+
+    @app.get("/admin/trace")
+    async def trace_tail(n: int = 100, request_id: Optional[str] = None):
+        """Tail of the per-request span log (arrived/routed/finished)."""
+        return gw.spans.tail(n=n, request_id=request_id)
 
     @app.post("/admin/drain/{engine_name}")
     async def drain_engine(engine_name: str, on: bool = True):
