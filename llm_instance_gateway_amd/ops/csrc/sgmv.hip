@@ -111,16 +111,140 @@ __global__ void sgmv_expand_kernel(
   }
 }
 
+// ---- large-T (prefill) variants -------------------------------------------
+//
+// At prefill shapes (T up to 4096) the v2 kernels become BANDWIDTH-bound on
+// redundant adapter-weight reads: shrink re-reads A per (token, row) — T*R*H
+// traffic; expand re-reads B per token — T*O*R.  Prefill batches are
+// sequence-contiguous, so consecutive tokens share an adapter: amortize the
+// weight reads across a token group.
+//
+// shrink_wide: one wave per (8 consecutive tokens, rank row).  Lane = 8
+//   token-subgroups x 8 H-stripes; the 8 subgroups read the SAME A chunk in
+//   the same cycle (one L1 fetch) -> A traffic / 8.
+// expand_wide: one thread per (4 consecutive tokens, 8 outputs).  B rows are
+//   loaded once per thread and reused for 4 tokens (per-token reload only on
+//   a slot boundary) -> B traffic / 4; y RMW traffic is the floor.
+
+__global__ void sgmv_shrink_wide_kernel(
+    uint16_t* __restrict__ u, const uint16_t* __restrict__ x,
+    const uint16_t* __restrict__ a, const int* __restrict__ slots,
+    const int T, const int R, const int H) {
+  // wave handles tokens [t8*8, t8*8+8) x rank row r
+  const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int nt8 = (T + 7) / 8;
+  const int t8 = wid / R;
+  if (t8 >= nt8) return;
+  const int r = wid - t8 * R;
+  const int lane = threadIdx.x & 63;
+  const int tg = lane >> 3;   // token subgroup 0..7
+  const int li8 = lane & 7;   // H stripe 0..7
+  const int t = t8 * 8 + tg;
+  const int tc = t < T ? t : T - 1;  // clamp reads; store is guarded
+  const int s = slots[tc];
+  const int chunks = H / 8;
+
+  float acc = 0.0f;
+  if (s >= 0) {
+    const uint16_t* xrow = x + (size_t)tc * H;
+    const uint16_t* arow = a + ((size_t)s * R + r) * H;
+#pragma unroll 4
+    for (int c = li8; c < chunks; c += 8) {
+      f32x8 xv = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(xrow + c * 8));
+      f32x8 av = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(arow + c * 8));
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc += xv[e] * av[e];
+    }
+  }
+  // reduce across the 8 H-stripes of this token subgroup
+#pragma unroll
+  for (int st = 4; st >= 1; st >>= 1)
+    acc += __shfl_xor(acc, st, 64);
+  if (li8 == 0 && t < T && s >= 0) u[(size_t)t * R + r] = f32_to_bf16(acc);
+}
+
+template <int R>
+__global__ void sgmv_expand_wide_kernel(
+    uint16_t* __restrict__ y, const uint16_t* __restrict__ u,
+    const uint16_t* __restrict__ b, const int* __restrict__ slots,
+    const float scaling, const int T, const int O) {
+  const int nt4 = (T + 3) / 4;
+  const long total = (long)nt4 * (O / 8);
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += gridDim.x * (long)blockDim.x) {
+    const int oc = idx % (O / 8);
+    const long t4 = idx / (O / 8);
+    const long t0 = t4 * 4;
+    int cur_slot = -2;  // force first load
+    u16x8 breg[8 * (R / 8)];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const long t = t0 + j;
+      if (t >= T) break;
+      const int s = slots[t];
+      if (s < 0) continue;
+      if (s != cur_slot) {
+        cur_slot = s;
+        const uint16_t* brow = b + ((size_t)s * O + oc * 8) * R;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+#pragma unroll
+          for (int rc = 0; rc < R / 8; ++rc)
+            breg[e * (R / 8) + rc] = *reinterpret_cast<const u16x8*>(
+                brow + (size_t)e * R + rc * 8);
+      }
+      float uv[R];
+#pragma unroll
+      for (int rc = 0; rc < R / 8; ++rc) {
+        f32x8 v = bf16x8_to_f32x8(
+            *reinterpret_cast<const u16x8*>(u + (size_t)t * R + rc * 8));
+#pragma unroll
+        for (int e = 0; e < 8; ++e) uv[rc * 8 + e] = v[e];
+      }
+      u16x8 yv = *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8);
+      f32x8 acc;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] = bf16_to_f32(yv[e]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float aa = 0.0f;
+#pragma unroll
+        for (int rc = 0; rc < R / 8; ++rc) {
+          f32x8 bv = bf16x8_to_f32x8(breg[e * (R / 8) + rc]);
+#pragma unroll
+          for (int q = 0; q < 8; ++q) aa += bv[q] * uv[rc * 8 + q];
+        }
+        acc[e] += aa * scaling;
+      }
+      *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8) =
+          f32x8_to_bf16x8(acc);
+    }
+  }
+}
+
+// v2 kernels above stay the decode path: at small T they put more waves in
+// flight per unit work; the _wide variants win once weight-read amortization
+// dominates (threshold measured on hardware, benchmarks/bench_sgmv.py).
+#define SGMV_WIDE_MIN_T 256
+
 extern "C" {
 
 void lig_sgmv_shrink_launch(uint16_t* u, const uint16_t* x, const uint16_t* a,
                             const int* slots, int T, int R, int H,
                             hipStream_t stream) {
-  // 4 waves per block, one wave per (token, rank-row)
+  dim3 block(256);
+  if (T >= SGMV_WIDE_MIN_T) {
+    const long waves = (long)((T + 7) / 8) * R;
+    long grid = (waves + 3) / 4;
+    if (grid > 1048576) grid = 1048576;
+    hipLaunchKernelGGL(sgmv_shrink_wide_kernel, dim3((int)grid), block, 0,
+                       stream, u, x, a, slots, T, R, H);
+    return;
+  }
+  // one wave per (token, rank-row)
   const long waves = (long)T * R;
   long grid = (waves + 3) / 4;
   if (grid > 1048576) grid = 1048576;
-  dim3 block(256);
   hipLaunchKernelGGL(sgmv_shrink_kernel, dim3((int)grid), block, 0, stream, u,
                      x, a, slots, T, R, H);
 }
@@ -128,13 +252,21 @@ void lig_sgmv_shrink_launch(uint16_t* u, const uint16_t* x, const uint16_t* a,
 void lig_sgmv_expand_launch(uint16_t* y, const uint16_t* u, const uint16_t* b,
                             const int* slots, float scaling, int T, int R,
                             int O, hipStream_t stream) {
-  const long total = (long)T * (O / 8);
+  const bool wide = T >= SGMV_WIDE_MIN_T;
+  const long total = wide ? (long)((T + 3) / 4) * (O / 8) : (long)T * (O / 8);
   int grid = (int)((total + 255) / 256);
   if (grid > 65535) grid = 65535;
   if (grid < 1) grid = 1;
-#define LAUNCH_E(RR)                                                       \
-  hipLaunchKernelGGL((sgmv_expand_kernel<RR>), dim3(grid), dim3(256), 0,   \
-                     stream, y, u, b, slots, scaling, T, O)
+#define LAUNCH_E(RR)                                                        \
+  do {                                                                      \
+    if (wide)                                                               \
+      hipLaunchKernelGGL((sgmv_expand_wide_kernel<RR>), dim3(grid),         \
+                         dim3(256), 0, stream, y, u, b, slots, scaling, T,  \
+                         O);                                                \
+    else                                                                    \
+      hipLaunchKernelGGL((sgmv_expand_kernel<RR>), dim3(grid), dim3(256),   \
+                         0, stream, y, u, b, slots, scaling, T, O);         \
+  } while (0)
   switch (R) {
     case 8: LAUNCH_E(8); break;
     case 16: LAUNCH_E(16); break;

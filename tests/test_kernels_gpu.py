@@ -166,7 +166,15 @@ def test_engine_e2e_gpu_matches_cpu_greedy():
     assert agree >= total * 0.7, f"only {agree}/{total} tokens agree"
 
 
-@pytest.mark.parametrize("T,H,R,O", [(64, 4096, 16, 6144), (3, 64, 8, 128)])
+@pytest.mark.parametrize(
+    "T,H,R,O",
+    [
+        (64, 4096, 16, 6144),   # decode path (v2 kernels)
+        (3, 64, 8, 128),        # tiny/ragged
+        (517, 4096, 16, 6144),  # wide path (T >= 256), non-multiple of 4/8
+        (1024, 4096, 32, 4096),  # wide, higher rank
+    ],
+)
 def test_sgmv_kernels(T, H, R, O):
     torch.manual_seed(9)
     S = 4
