@@ -95,6 +95,11 @@ def main() -> None:
     else:
         device = "cpu"
 
+    if device.startswith("cuda"):
+        from llm_instance_gateway_amd.utils.tunableop import maybe_enable_tunableop
+
+        maybe_enable_tunableop()
+
     mc = get_model_config(args.model)
     is_cuda = device.startswith("cuda")
     cfg = EngineConfig(

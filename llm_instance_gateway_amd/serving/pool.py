@@ -146,6 +146,9 @@ def _worker_main(
 
     if device.startswith("cuda"):
         torch.cuda.set_device(device)
+        from ..utils.tunableop import maybe_enable_tunableop
+
+        maybe_enable_tunableop()
     tp = None
     if tp_size > 1:
         from ..parallel import TPGroup, init_distributed
