@@ -17,7 +17,7 @@ sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="llama3-8b")
-    ap.add_argument("--steps", type=int, default=6)
+    ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--concurrency", type=int, default=128)
     args = ap.parse_args()
 
@@ -44,11 +44,14 @@ def main():
     torch.manual_seed(0)
     eng = LLMEngine(cfg, device="cuda:0")
     g = torch.Generator().manual_seed(1)
+    # short prompts: a couple of prefill steps, then a long run of M=concurrency
+    # decode steps — the decode GEMM shapes (M = graph bucket = concurrency)
+    # are the stable, highest-leverage shapes to tune
     for i in range(args.concurrency):
-        plen = int(torch.randint(64, 512, (1,), generator=g))
+        plen = int(torch.randint(8, 24, (1,), generator=g))
         eng.add_request(
             torch.randint(1, 128000, (plen,), generator=g).tolist(),
-            SamplingParams(max_new_tokens=32),
+            SamplingParams(max_new_tokens=args.steps + 8),
         )
     for _ in range(args.steps):
         if not eng.has_work():
