@@ -1,0 +1,72 @@
+#!/usr/bin/env python3
+"""Paged-attention kernel micro-benchmark + numerics gate (GPU).
+
+Decode: T sequences x 1 token, context sweep.  Prefill: chunked varlen.
+
+    python benchmarks/bench_attention.py
+"""
+
+import sys
+import time
+from pathlib import Path
+
+import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+from llm_instance_gateway_amd import ops
+from llm_instance_gateway_amd.ops import torch_ref
+
+BS = 16  # block size
+HKV, HQ, D = 8, 32, 128
+
+
+def make_cache(num_blocks, dev):
+    k = torch.randn(num_blocks + 1, HKV, BS, D, dtype=torch.bfloat16, device=dev) * 0.2
+    v = torch.randn(num_blocks + 1, HKV, BS, D, dtype=torch.bfloat16, device=dev) * 0.2
+    return k, v
+
+
+def bench_decode(T, ctx, iters=200):
+    dev = "cuda:0"
+    torch.manual_seed(0)
+    blocks_per_seq = (ctx + BS - 1) // BS
+    nb = T * blocks_per_seq
+    k, v = make_cache(nb, dev)
+    q = torch.randn(T, HQ, D, dtype=torch.bfloat16, device=dev) * 0.3
+    bt = torch.arange(nb, dtype=torch.int32, device=dev).reshape(T, blocks_per_seq)
+    ctx_t = torch.full((T,), ctx, dtype=torch.int32, device=dev)
+    scale = D ** -0.5
+
+    out = ops.paged_attention_decode(q, k, v, bt, ctx_t, scale)
+    ref = torch_ref.paged_attention_decode(
+        q.float(), k.float(), v.float(), bt, ctx_t, scale
+    )
+    err = (out.float() - ref).abs().max().item()
+    assert err < 0.05, f"numerics: {err}"
+
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ops.paged_attention_decode(q, k, v, bt, ctx_t, scale)
+    torch.cuda.synchronize()
+    us = (time.perf_counter() - t0) / iters * 1e6
+    kv_mb = T * ctx * HKV * D * 2 * 2 / 1e6
+    print(
+        f"decode T={T:4d} ctx={ctx:5d}: {us:7.1f} us  "
+        f"({kv_mb:.0f} MB KV -> {kv_mb / us * 1e6 / 1e12:.2f} TB/s effective)"
+    )
+    return us
+
+
+def main():
+    assert torch.cuda.is_available() and ops.native_available()
+    for ctx in (128, 256, 512, 1024, 2048):
+        bench_decode(128, ctx)
+    bench_decode(256, 384)
+    bench_decode(32, 384)
+    print("OK")
+
+
+if __name__ == "__main__":
+    main()

@@ -80,34 +80,60 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
   const int* bt = block_tables + (size_t)seq * max_blocks;
   const int iters = (ctx + 7) / 8;
 
-  for (int it = 0; it < iters; ++it) {
+  // Software-pipelined K/V: the next iteration's 4 vector loads are issued
+  // BEFORE this iteration's dot/softmax/accumulate chain (whose xor-shuffle
+  // reduces serialize ~10 cross-lane ops), so HBM latency hides behind
+  // compute instead of stalling every iteration.
+  constexpr int NV = DL >= 8 ? DL / 8 : 1;  // u16x8 vectors per lane
+  u16x8 kraw[2][NV], vraw[2][NV];
+  bool vbuf[2] = {false, false};
+
+  auto issue_load = [&](int buf, int it) {
     const int t = it * 8 + tok;
     const bool valid = t < ctx;
+    vbuf[buf] = valid;
+    if (!valid) return;
+    const int blk = bt[t / block_size];
+    const int off = t % block_size;
+    const size_t base =
+        (((size_t)blk * num_kv_heads + kvh) * block_size + off) * HEAD_DIM +
+        dchunk * DL;
+    if constexpr (DL >= 8) {
+#pragma unroll
+      for (int i = 0; i < NV; ++i) {
+        kraw[buf][i] =
+            *reinterpret_cast<const u16x8*>(k_cache + base + i * 8);
+        vraw[buf][i] =
+            *reinterpret_cast<const u16x8*>(v_cache + base + i * 8);
+      }
+    } else {  // small head_dim: exactly DL scalars (no row overrun)
+#pragma unroll
+      for (int i = 0; i < DL; ++i) {
+        kraw[buf][0][i] = k_cache[base + i];
+        vraw[buf][0][i] = v_cache[base + i];
+      }
+    }
+  };
+
+  if (iters > 0) issue_load(0, 0);
+
+  for (int it = 0; it < iters; ++it) {
+    const int cur = it & 1;
+    if (it + 1 < iters) issue_load(cur ^ 1, it + 1);
+    const bool valid = vbuf[cur];
+
     float kf[DL];
-    const uint16_t* krow = nullptr;
-    const uint16_t* vrow = nullptr;
-    if (valid) {
-      const int blk = bt[t / block_size];
-      const int off = t % block_size;
-      const size_t base =
-          (((size_t)blk * num_kv_heads + kvh) * block_size + off) * HEAD_DIM +
-          dchunk * DL;
-      krow = k_cache + base;
-      vrow = v_cache + base;
-      if constexpr (DL >= 8) {
+    if constexpr (DL >= 8) {
 #pragma unroll
-        for (int i = 0; i < DL; i += 8) {
-          f32x8 v8 = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(krow + i));
+      for (int i = 0; i < NV; ++i) {
+        f32x8 v8 = bf16x8_to_f32x8(kraw[cur][i]);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) kf[i + j] = v8[j];
-        }
-      } else {
-#pragma unroll
-        for (int i = 0; i < DL; ++i) kf[i] = bf16_to_f32(krow[i]);
+        for (int j = 0; j < 8; ++j) kf[i * 8 + j] = valid ? v8[j] : 0.0f;
       }
     } else {
 #pragma unroll
-      for (int i = 0; i < DL; ++i) kf[i] = 0.0f;
+      for (int i = 0; i < DL; ++i)
+        kf[i] = valid ? bf16_to_f32(kraw[cur][0][i]) : 0.0f;
     }
 
     float s = 0.0f;
@@ -131,21 +157,17 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
     l = l * alpha + stride_reduce_sum<8, 8>(p);
 
     float vf[DL];
-    if (valid) {
-      if constexpr (DL >= 8) {
+    if constexpr (DL >= 8) {
 #pragma unroll
-        for (int i = 0; i < DL; i += 8) {
-          f32x8 v8 = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(vrow + i));
+      for (int i = 0; i < NV; ++i) {
+        f32x8 v8 = bf16x8_to_f32x8(vraw[cur][i]);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) vf[i + j] = v8[j];
-        }
-      } else {
-#pragma unroll
-        for (int i = 0; i < DL; ++i) vf[i] = bf16_to_f32(vrow[i]);
+        for (int j = 0; j < 8; ++j) vf[i * 8 + j] = valid ? v8[j] : 0.0f;
       }
     } else {
 #pragma unroll
-      for (int i = 0; i < DL; ++i) vf[i] = 0.0f;
+      for (int i = 0; i < DL; ++i)
+        vf[i] = valid ? bf16_to_f32(vraw[cur][0][i]) : 0.0f;
     }
 #pragma unroll
     for (int i = 0; i < DL; ++i) oacc[i] = oacc[i] * alpha + p * vf[i];
