@@ -33,6 +33,11 @@ def maybe_enable_tunableop() -> bool:
     if not torch.cuda.is_available():
         return False
     tuning = os.environ.get("LIG_TUNE_GEMMS", "0") == "1"
+    # MEASURED on MI355X (gpurun_out/bench4.log): enabling TunableOp read-only
+    # during serving made the bench 3x SLOWER (35.8 vs 12.4 ms/step) — the
+    # tunable GEMM wrapper defeats hipGraph decode capture.  Strictly opt-in.
+    if not tuning and os.environ.get("LIG_TUNABLEOP", "0") != "1":
+        return False
     if not tuning and not RESULTS_FILE.exists():
         return False
     import torch.cuda.tunable as tunable
