@@ -11,7 +11,9 @@ zero staleness.
 
 from __future__ import annotations
 
+import atexit
 import logging
+import os
 import time
 from typing import Callable, Dict, List, Optional
 
@@ -24,6 +26,20 @@ from .scheduler import EngineScheduler, SchedulerOutput
 from .sequence import SamplingParams, Sequence, SequenceStatus
 
 log = logging.getLogger(__name__)
+
+# LIG_STEP_TIMING=1: accumulate per-step host/GPU phase times (schedule+lora
+# vs execute-with-sync), dumped at exit — the host-gap diagnosis knob.
+_STEP_TIMING = os.environ.get("LIG_STEP_TIMING", "0") == "1"
+_STEP_STATS: Dict[str, list] = {}
+
+if _STEP_TIMING:  # pragma: no cover - diagnostic only
+    @atexit.register
+    def _dump_step_stats() -> None:
+        for kind, (n, sched_s, exec_s) in sorted(_STEP_STATS.items()):
+            print(
+                f"[step-timing] {kind}: n={n} schedule+lora={sched_s / max(1, n) * 1e3:.2f}ms "
+                f"execute(sync)={exec_s / max(1, n) * 1e3:.2f}ms"
+            )
 
 
 class LLMEngine:
@@ -80,6 +96,8 @@ class LLMEngine:
     def step(self) -> List[Sequence]:
         """One continuous-batching iteration.  Returns sequences finished in
         this step."""
+        timing = _STEP_TIMING
+        t0 = time.perf_counter() if timing else 0.0
         out: SchedulerOutput = self.scheduler.schedule()
         if out.is_empty:
             return []
@@ -90,7 +108,17 @@ class LLMEngine:
                 [ss.seq.lora_name for ss in out.scheduled if ss.seq.lora_name]
             )
             self.runner.adapter_manager = self.adapter_manager
+        if timing:
+            t1 = time.perf_counter()
         sampled = self.runner.execute(out)
+        if timing:
+            torch.cuda.synchronize() if self.runner.device.type == "cuda" else None
+            t2 = time.perf_counter()
+            kind = "prefill" if any(ss.num_tokens > 1 for ss in out.scheduled) else "decode"
+            acc = _STEP_STATS.setdefault(kind, [0, 0.0, 0.0])
+            acc[0] += 1
+            acc[1] += t1 - t0
+            acc[2] += t2 - t1
         for ss in out.scheduled:
             seq = ss.seq
             tok = sampled.get(seq.seq_id)
