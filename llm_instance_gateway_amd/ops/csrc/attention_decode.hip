@@ -80,98 +80,96 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
   const int* bt = block_tables + (size_t)seq * max_blocks;
   const int iters = (ctx + 7) / 8;
 
-  // Software-pipelined K/V: the next iteration's 4 vector loads are issued
+  // Software-pipelined K/V: the next iteration's vector loads are issued
   // BEFORE this iteration's dot/softmax/accumulate chain (whose xor-shuffle
   // reduces serialize ~10 cross-lane ops), so HBM latency hides behind
-  // compute instead of stalling every iteration.
+  // compute.  The two prefetch buffers are SEPARATE NAMED arrays selected at
+  // compile time by a manually 2-unrolled loop: a runtime-indexed buf[2][NV]
+  // spills to scratch (first attempt measured 6x slower than no prefetch).
   constexpr int NV = DL >= 8 ? DL / 8 : 1;  // u16x8 vectors per lane
-  u16x8 kraw[2][NV], vraw[2][NV];
-  bool vbuf[2] = {false, false};
+  u16x8 kraw0[NV], vraw0[NV], kraw1[NV], vraw1[NV];
+  bool valid0 = false, valid1 = false;
 
-  auto issue_load = [&](int buf, int it) {
-    const int t = it * 8 + tok;
-    const bool valid = t < ctx;
-    vbuf[buf] = valid;
-    if (!valid) return;
-    const int blk = bt[t / block_size];
-    const int off = t % block_size;
-    const size_t base =
-        (((size_t)blk * num_kv_heads + kvh) * block_size + off) * HEAD_DIM +
-        dchunk * DL;
-    if constexpr (DL >= 8) {
-#pragma unroll
-      for (int i = 0; i < NV; ++i) {
-        kraw[buf][i] =
-            *reinterpret_cast<const u16x8*>(k_cache + base + i * 8);
-        vraw[buf][i] =
-            *reinterpret_cast<const u16x8*>(v_cache + base + i * 8);
-      }
-    } else {  // small head_dim: exactly DL scalars (no row overrun)
-#pragma unroll
-      for (int i = 0; i < DL; ++i) {
-        kraw[buf][0][i] = k_cache[base + i];
-        vraw[buf][0][i] = v_cache[base + i];
-      }
-    }
-  };
+#define ISSUE_LOAD(KBUF, VBUF, VALID, IT)                                    \
+  do {                                                                       \
+    const int t_ = (IT) * 8 + tok;                                           \
+    (VALID) = t_ < ctx;                                                      \
+    if (VALID) {                                                             \
+      const int blk_ = bt[t_ / block_size];                                  \
+      const size_t base_ = (((size_t)blk_ * num_kv_heads + kvh) *            \
+                                block_size + (t_ % block_size)) * HEAD_DIM + \
+                           dchunk * DL;                                      \
+      if constexpr (DL >= 8) {                                               \
+        _Pragma("unroll") for (int i_ = 0; i_ < NV; ++i_) {                  \
+          (KBUF)[i_] =                                                       \
+              *reinterpret_cast<const u16x8*>(k_cache + base_ + i_ * 8);     \
+          (VBUF)[i_] =                                                       \
+              *reinterpret_cast<const u16x8*>(v_cache + base_ + i_ * 8);     \
+        }                                                                    \
+      } else {                                                               \
+        _Pragma("unroll") for (int i_ = 0; i_ < DL; ++i_) {                  \
+          (KBUF)[0][i_] = k_cache[base_ + i_];                               \
+          (VBUF)[0][i_] = v_cache[base_ + i_];                               \
+        }                                                                    \
+      }                                                                      \
+    }                                                                        \
+  } while (0)
 
-  if (iters > 0) issue_load(0, 0);
+#define STEP_BODY(KBUF, VBUF, VALID)                                         \
+  do {                                                                       \
+    const bool valid = (VALID);                                              \
+    float kf[DL];                                                            \
+    if constexpr (DL >= 8) {                                                 \
+      _Pragma("unroll") for (int i = 0; i < NV; ++i) {                       \
+        f32x8 v8 = bf16x8_to_f32x8((KBUF)[i]);                               \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j)                        \
+            kf[i * 8 + j] = valid ? v8[j] : 0.0f;                            \
+      }                                                                      \
+    } else {                                                                 \
+      _Pragma("unroll") for (int i = 0; i < DL; ++i)                         \
+          kf[i] = valid ? bf16_to_f32((KBUF)[0][i]) : 0.0f;                  \
+    }                                                                        \
+    float s = 0.0f;                                                          \
+    _Pragma("unroll") for (int i = 0; i < DL; ++i) s += qreg[i] * kf[i];     \
+    s = group_reduce_sum<8>(s);                                              \
+    s = valid ? s * scale : -INFINITY;                                       \
+    const float chunk_max = stride_reduce_max<8, 8>(s);                      \
+    const float new_m = fmaxf(m, chunk_max);                                 \
+    float p, alpha;                                                          \
+    if (new_m == -INFINITY) {                                                \
+      p = 0.0f;                                                              \
+      alpha = 1.0f;                                                          \
+    } else {                                                                 \
+      alpha = (m == -INFINITY) ? 0.0f : __expf(m - new_m);                   \
+      p = (s == -INFINITY) ? 0.0f : __expf(s - new_m);                       \
+      m = new_m;                                                             \
+    }                                                                        \
+    l = l * alpha + stride_reduce_sum<8, 8>(p);                              \
+    float vf[DL];                                                            \
+    if constexpr (DL >= 8) {                                                 \
+      _Pragma("unroll") for (int i = 0; i < NV; ++i) {                       \
+        f32x8 v8 = bf16x8_to_f32x8((VBUF)[i]);                               \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j)                        \
+            vf[i * 8 + j] = valid ? v8[j] : 0.0f;                            \
+      }                                                                      \
+    } else {                                                                 \
+      _Pragma("unroll") for (int i = 0; i < DL; ++i)                         \
+          vf[i] = valid ? bf16_to_f32((VBUF)[0][i]) : 0.0f;                  \
+    }                                                                        \
+    _Pragma("unroll") for (int i = 0; i < DL; ++i)                           \
+        oacc[i] = oacc[i] * alpha + p * vf[i];                               \
+  } while (0)
 
-  for (int it = 0; it < iters; ++it) {
-    const int cur = it & 1;
-    if (it + 1 < iters) issue_load(cur ^ 1, it + 1);
-    const bool valid = vbuf[cur];
-
-    float kf[DL];
-    if constexpr (DL >= 8) {
-#pragma unroll
-      for (int i = 0; i < NV; ++i) {
-        f32x8 v8 = bf16x8_to_f32x8(kraw[cur][i]);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) kf[i * 8 + j] = valid ? v8[j] : 0.0f;
-      }
-    } else {
-#pragma unroll
-      for (int i = 0; i < DL; ++i)
-        kf[i] = valid ? bf16_to_f32(kraw[cur][0][i]) : 0.0f;
-    }
-
-    float s = 0.0f;
-#pragma unroll
-    for (int i = 0; i < DL; ++i) s += qreg[i] * kf[i];
-    s = group_reduce_sum<8>(s);  // full dot, replicated in the token's 8 lanes
-    s = valid ? s * scale : -INFINITY;
-
-    const float chunk_max = stride_reduce_max<8, 8>(s);
-    const float new_m = fmaxf(m, chunk_max);
-    float p;
-    float alpha;
-    if (new_m == -INFINITY) {  // whole iteration masked (cannot happen mid-seq)
-      p = 0.0f;
-      alpha = 1.0f;
-    } else {
-      alpha = (m == -INFINITY) ? 0.0f : __expf(m - new_m);
-      p = (s == -INFINITY) ? 0.0f : __expf(s - new_m);
-      m = new_m;
-    }
-    l = l * alpha + stride_reduce_sum<8, 8>(p);
-
-    float vf[DL];
-    if constexpr (DL >= 8) {
-#pragma unroll
-      for (int i = 0; i < NV; ++i) {
-        f32x8 v8 = bf16x8_to_f32x8(vraw[cur][i]);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vf[i * 8 + j] = valid ? v8[j] : 0.0f;
-      }
-    } else {
-#pragma unroll
-      for (int i = 0; i < DL; ++i)
-        vf[i] = valid ? bf16_to_f32(vraw[cur][0][i]) : 0.0f;
-    }
-#pragma unroll
-    for (int i = 0; i < DL; ++i) oacc[i] = oacc[i] * alpha + p * vf[i];
+  if (iters > 0) ISSUE_LOAD(kraw0, vraw0, valid0, 0);
+  for (int it = 0; it < iters; it += 2) {
+    if (it + 1 < iters) ISSUE_LOAD(kraw1, vraw1, valid1, it + 1);
+    STEP_BODY(kraw0, vraw0, valid0);
+    if (it + 1 >= iters) break;
+    if (it + 2 < iters) ISSUE_LOAD(kraw0, vraw0, valid0, it + 2);
+    STEP_BODY(kraw1, vraw1, valid1);
   }
+#undef ISSUE_LOAD
+#undef STEP_BODY
 
   // combine the 8 token-subsets (lanes stride-8 apart share dchunk)
 #pragma unroll
