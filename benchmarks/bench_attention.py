@@ -54,7 +54,7 @@ def bench_decode(T, ctx, iters=200):
     kv_mb = T * ctx * HKV * D * 2 * 2 / 1e6
     print(
         f"decode T={T:4d} ctx={ctx:5d}: {us:7.1f} us  "
-        f"({kv_mb:.0f} MB KV -> {kv_mb / us * 1e6 / 1e12:.2f} TB/s effective)"
+        f"({kv_mb:.0f} MB KV -> {kv_mb / us:.2f} TB/s effective)"
     )
     return us
 
