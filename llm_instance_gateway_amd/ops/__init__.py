@@ -214,3 +214,27 @@ def sgmv_expand(u, b_stack, slots, y, scaling=1.0):
         _EXT.sgmv_expand(y, u, b_stack, slots, float(scaling), _stream())
         return
     torch_ref.sgmv_expand(u, b_stack, slots, y, scaling)
+
+
+# Decode-batch projection GEMMs go through the hand-written MFMA kernel when
+# hipBLASLt's skinny-M heuristics underperform (see ops/csrc/skinny_gemm.hip).
+SKINNY_GEMM_MAX_M = 256
+
+
+def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """y = x @ weight.T (nn.Linear, no bias).  Dispatch: custom MFMA kernel
+    for decode-sized M on GPU; library GEMM otherwise."""
+    if (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and 0 < x.shape[0] <= SKINNY_GEMM_MAX_M
+        and weight.shape[1] % 32 == 0
+        and weight.shape[0] % 64 == 0
+        and _use_native(x)
+    ):
+        out = torch.empty(
+            x.shape[0], weight.shape[0], dtype=x.dtype, device=x.device
+        )
+        _EXT.skinny_gemm(out, x.contiguous(), weight, _stream())
+        return out
+    return torch.nn.functional.linear(x, weight)

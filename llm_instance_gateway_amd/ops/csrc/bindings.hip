@@ -32,6 +32,8 @@ void lig_sgmv_shrink_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             const int*, int, int, int, hipStream_t);
 void lig_sgmv_expand_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             const int*, float, int, int, int, hipStream_t);
+void lig_skinny_gemm_launch(uint16_t*, const uint16_t*, const uint16_t*,
+                            int, int, int, hipStream_t);
 void lig_paged_attention_prefill_launch(
     uint16_t*, const uint16_t*, const uint16_t*, const uint16_t*, const int*,
     const int*, const int*, const int*, const int*, float, int, int, int,
@@ -198,6 +200,20 @@ void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
                          slots.data_ptr<int>(), (float)scaling, T, R, O, st);
 }
 
+void skinny_gemm(torch::Tensor c, torch::Tensor a, torch::Tensor w,
+                 int64_t stream) {
+  const int M = a.size(0);
+  const int K = a.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && c.size(0) == M && c.size(1) == N,
+              "skinny_gemm shape mismatch");
+  TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "skinny_gemm needs K%32==0, N%64==0");
+  TORCH_CHECK(a.is_contiguous() && w.is_contiguous() && c.is_contiguous());
+  if (M == 0) return;
+  lig_skinny_gemm_launch(bf16_ptr(c), bf16_cptr(a), bf16_cptr(w), M, N, K,
+                         as_stream(stream));
+}
+
 void sgmv_expand(torch::Tensor y, torch::Tensor u, torch::Tensor b_stack,
                  torch::Tensor slots, double scaling, int64_t stream) {
   TORCH_CHECK(slots.scalar_type() == torch::kInt32);
@@ -225,6 +241,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_prefill", &paged_attention_prefill);
   m.def("sgmv_shrink", &sgmv_shrink);
   m.def("sgmv_apply", &sgmv_apply);
+  m.def("skinny_gemm", &skinny_gemm);
   m.def("sgmv_expand", &sgmv_expand);
   m.def("mfma_probe", [](torch::Tensor out, torch::Tensor a, torch::Tensor b,
                          int64_t stream) {
