@@ -67,7 +67,7 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=128)
     ap.add_argument("--warmup", type=int, default=32)
     ap.add_argument("--model", default="llama3-8b")
-    ap.add_argument("--concurrency", type=int, default=128)
+    ap.add_argument("--concurrency", type=int, default=256)
     ap.add_argument("--device", default=None, help="override (e.g. cpu for dev)")
     ap.add_argument(
         "--lora",
