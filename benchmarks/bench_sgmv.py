@@ -19,19 +19,27 @@ from llm_instance_gateway_amd import ops
 from llm_instance_gateway_amd.ops import torch_ref
 
 
-def bench_case(T, H, O, R, S=32, iters=200):
+def bench_case(T, H, O, R, S=32, iters=200, wide=False):
     dev = "cuda:0"
     torch.manual_seed(0)
     x = torch.randn(T, H, device=dev, dtype=torch.bfloat16) * 0.1
     a = torch.randn(S, R, H, device=dev, dtype=torch.bfloat16) * 0.02
     b = torch.randn(S, O, R, device=dev, dtype=torch.bfloat16) * 0.02
-    slots = torch.randint(0, S, (T,), device=dev, dtype=torch.int32)
+    if wide:
+        # prefill regime: sequence-contiguous adapter segments (~200 tokens)
+        seg = []
+        g = torch.Generator().manual_seed(1)
+        while len(seg) < T:
+            seg += [int(torch.randint(0, S, (1,), generator=g))] * 200
+        slots = torch.tensor(seg[:T], device=dev, dtype=torch.int32)
+    else:
+        slots = torch.randint(0, S, (T,), device=dev, dtype=torch.int32)
     y = torch.randn(T, O, device=dev, dtype=torch.bfloat16) * 0.1
     u_ws = torch.zeros(T, R, device=dev, dtype=torch.bfloat16)
 
     # numerics vs fp32 reference
     y_test = y.clone()
-    ops._EXT.sgmv_apply(y_test, x, a, b, u_ws, slots, 1.0, torch.cuda.current_stream().cuda_stream)
+    ops._EXT.sgmv_apply(y_test, x, a, b, u_ws, slots, 1.0, wide, torch.cuda.current_stream().cuda_stream)
     torch.cuda.synchronize()
     u_ref = torch_ref.sgmv_shrink(x.float(), a.float(), slots)
     y_ref = y.clone().float()
@@ -42,11 +50,11 @@ def bench_case(T, H, O, R, S=32, iters=200):
 
     stream = torch.cuda.current_stream().cuda_stream
     for _ in range(20):
-        ops._EXT.sgmv_apply(y, x, a, b, u_ws, slots, 1.0, stream)
+        ops._EXT.sgmv_apply(y, x, a, b, u_ws, slots, 1.0, wide, stream)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        ops._EXT.sgmv_apply(y, x, a, b, u_ws, slots, 1.0, stream)
+        ops._EXT.sgmv_apply(y, x, a, b, u_ws, slots, 1.0, wide, stream)
     torch.cuda.synchronize()
     us = (time.perf_counter() - t0) / iters * 1e6
     gb = (T * H * 2 + T * R * 4 + S * R * H * 2 / 8 + T * O * 4 + T * R * O / O * 2) / 1e9
@@ -63,9 +71,10 @@ def main():
     print("== decode shapes (qkv / o / gate_up / down of llama3-8b) ==")
     for (H, O) in ((4096, 6144), (4096, 4096), (4096, 28672), (14336, 4096)):
         bench_case(128, H, O, 16)
-    print("== prefill shape ==")
-    bench_case(512, 4096, 6144, 16)
-    bench_case(2048, 4096, 6144, 16)
+    print("== prefill shapes (segmented slots, wide kernels) ==")
+    bench_case(512, 4096, 6144, 16, wide=True)
+    bench_case(2048, 4096, 6144, 16, wide=True)
+    bench_case(2048, 4096, 28672, 16, wide=True)
     print("== rank sweep (qkv) ==")
     for r in (8, 32, 64):
         bench_case(128, 4096, 6144, r)

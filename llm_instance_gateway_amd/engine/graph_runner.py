@@ -99,12 +99,18 @@ class DecodeGraphRunner:
         self.slot_mapping.fill_(self.scratch_block * self.block_size)
         self.lora_slots.fill_(-1)
         torch.cuda.synchronize(self.device)
-        # warmup on a side stream (torch.cuda.graph requirement)
+        # Warmup on a side stream (torch.cuda.graph requirement) — at EVERY
+        # capture batch size: hipBLASLt picks its algorithm + allocates its
+        # workspace on the FIRST call of each GEMM shape, and a first call
+        # that happens inside graph capture is restricted to slow no-workspace
+        # fallbacks which then get baked into the graph (measured: in-graph
+        # decode GEMMs 2-8x slower than the same shapes eager —
+        # profiles/r01_c256_breakdown.md).
         s = torch.cuda.Stream(device=self.device)
         s.wait_stream(torch.cuda.current_stream(self.device))
         with torch.cuda.stream(s):
-            for _ in range(2):
-                self._forward(min(8, self.max_bs))
+            for B in self.sizes:
+                self._forward(B)
         torch.cuda.current_stream(self.device).wait_stream(s)
         torch.cuda.synchronize(self.device)
         for B in self.sizes:

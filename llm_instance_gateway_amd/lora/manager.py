@@ -278,9 +278,12 @@ class AdapterManager:
                     max(T, 4096), self.rank, dtype=self.dtype, device=self.device
                 )
                 self._u_ws = ws
+            # wide kernels only for prefill batches (sequence-contiguous
+            # adapter segments); decode batches keep the per-token kernels
+            wide = bool(getattr(meta, "num_prefill_tokens", 0) > 0 and x.shape[0] >= 256)
             ops._EXT.sgmv_apply(
                 y, x, self.a_stacks[key], self.b_stacks[key], ws, slots, 1.0,
-                ops._stream(),
+                wide, ops._stream(),
             )
         else:
             u = ops.sgmv_shrink(x, self.a_stacks[key], slots)

@@ -29,9 +29,10 @@ void lig_paged_attention_decode_launch(uint16_t*, const uint16_t*,
 void lig_mfma_probe_launch(float*, const uint16_t*, const uint16_t*,
                            hipStream_t);
 void lig_sgmv_shrink_launch(uint16_t*, const uint16_t*, const uint16_t*,
-                            const int*, int, int, int, hipStream_t);
+                            const int*, int, int, int, int, hipStream_t);
 void lig_sgmv_expand_launch(uint16_t*, const uint16_t*, const uint16_t*,
-                            const int*, float, int, int, int, hipStream_t);
+                            const int*, float, int, int, int, int,
+                            hipStream_t);
 void lig_skinny_gemm_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             int, int, int, hipStream_t);
 void lig_paged_attention_prefill_launch(
@@ -177,14 +178,16 @@ void sgmv_shrink(torch::Tensor u, torch::Tensor x, torch::Tensor a_stack,
   TORCH_CHECK(H % 8 == 0 && R <= 64);
   if (T == 0) return;
   lig_sgmv_shrink_launch(bf16_ptr(u), bf16_cptr(x), bf16_cptr(a_stack),
-                         slots.data_ptr<int>(), T, R, H, as_stream(stream));
+                         slots.data_ptr<int>(), T, R, H, T >= 256,
+                         as_stream(stream));
 }
 
 // one call = shrink + expand (halves Python dispatch; u is a caller-owned
 // workspace so no per-call allocation)
 void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
                 torch::Tensor b_stack, torch::Tensor u_ws,
-                torch::Tensor slots, double scaling, int64_t stream) {
+                torch::Tensor slots, double scaling, bool use_wide,
+                int64_t stream) {
   TORCH_CHECK(slots.scalar_type() == torch::kInt32);
   const int T = x.size(0);
   const int H = x.size(1);
@@ -195,9 +198,10 @@ void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
   if (T == 0) return;
   hipStream_t st = as_stream(stream);
   lig_sgmv_shrink_launch(bf16_ptr(u_ws), bf16_cptr(x), bf16_cptr(a_stack),
-                         slots.data_ptr<int>(), T, R, H, st);
+                         slots.data_ptr<int>(), T, R, H, use_wide, st);
   lig_sgmv_expand_launch(bf16_ptr(y), bf16_cptr(u_ws), bf16_cptr(b_stack),
-                         slots.data_ptr<int>(), (float)scaling, T, R, O, st);
+                         slots.data_ptr<int>(), (float)scaling, T, R, O,
+                         use_wide, st);
 }
 
 void skinny_gemm(torch::Tensor c, torch::Tensor a, torch::Tensor w,
@@ -225,7 +229,7 @@ void sgmv_expand(torch::Tensor y, torch::Tensor u, torch::Tensor b_stack,
   if (T == 0) return;
   lig_sgmv_expand_launch(bf16_ptr(y), bf16_cptr(u), bf16_cptr(b_stack),
                          slots.data_ptr<int>(), (float)scaling, T, R, O,
-                         as_stream(stream));
+                         T >= 256, as_stream(stream));
 }
 
 }  // namespace

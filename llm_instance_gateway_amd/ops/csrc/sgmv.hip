@@ -222,18 +222,20 @@ __global__ void sgmv_expand_wide_kernel(
   }
 }
 
-// v2 kernels above stay the decode path: at small T they put more waves in
-// flight per unit work; the _wide variants win once weight-read amortization
-// dominates (threshold measured on hardware, benchmarks/bench_sgmv.py).
-#define SGMV_WIDE_MIN_T 256
+// v2 kernels above stay the decode path (mixed per-token slots: more waves
+// per unit work, no serial reload chains); the _wide variants win only on
+// PREFILL batches, whose tokens are sequence-contiguous so a token group
+// shares one adapter — the caller says which regime this is (`wide`).
+// Measured (profiles/r01_c256_breakdown.md): wide kernels on a 256-token
+// DECODE batch are 3-8x slower than v2 (per-token B reloads, half the waves).
 
 extern "C" {
 
 void lig_sgmv_shrink_launch(uint16_t* u, const uint16_t* x, const uint16_t* a,
-                            const int* slots, int T, int R, int H,
+                            const int* slots, int T, int R, int H, int wide,
                             hipStream_t stream) {
   dim3 block(256);
-  if (T >= SGMV_WIDE_MIN_T) {
+  if (wide) {
     const long waves = (long)((T + 7) / 8) * R;
     long grid = (waves + 3) / 4;
     if (grid > 1048576) grid = 1048576;
@@ -251,8 +253,7 @@ void lig_sgmv_shrink_launch(uint16_t* u, const uint16_t* x, const uint16_t* a,
 
 void lig_sgmv_expand_launch(uint16_t* y, const uint16_t* u, const uint16_t* b,
                             const int* slots, float scaling, int T, int R,
-                            int O, hipStream_t stream) {
-  const bool wide = T >= SGMV_WIDE_MIN_T;
+                            int O, int wide, hipStream_t stream) {
   const long total = wide ? (long)((T + 3) / 4) * (O / 8) : (long)T * (O / 8);
   int grid = (int)((total + 255) / 256);
   if (grid > 65535) grid = 65535;
