@@ -49,12 +49,36 @@ def bench_case(T, H, O, R, S=32, iters=200, wide=False):
     assert err <= 0.02 * max(1.0, scale), f"numerics off: {err} vs {scale}"
 
     stream = torch.cuda.current_stream().cuda_stream
+    if wide:
+        # MFMA tiled path (what the engine uses for prefill)
+        sl = slots.tolist()
+        t0s, lens, tslots = [], [], []
+        i = 0
+        while i < T:
+            j = i
+            while j < T and sl[j] == sl[i]:
+                j += 1
+            if sl[i] >= 0:
+                for off in range(i, j, 64):
+                    t0s.append(off)
+                    lens.append(min(64, j - off))
+                    tslots.append(sl[i])
+            i = j
+        tt0 = torch.tensor(t0s, dtype=torch.int32, device=dev)
+        tl = torch.tensor(lens, dtype=torch.int32, device=dev)
+        ts = torch.tensor(tslots, dtype=torch.int32, device=dev)
+
+        def run():
+            ops._EXT.sgmv_apply_tiled(y, x, a, b, u_ws, tt0, tl, ts, 1.0, stream)
+    else:
+        def run():
+            ops._EXT.sgmv_apply(y, x, a, b, u_ws, slots, 1.0, wide, stream)
     for _ in range(20):
-        ops._EXT.sgmv_apply(y, x, a, b, u_ws, slots, 1.0, wide, stream)
+        run()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        ops._EXT.sgmv_apply(y, x, a, b, u_ws, slots, 1.0, wide, stream)
+        run()
     torch.cuda.synchronize()
     us = (time.perf_counter() - t0) / iters * 1e6
     gb = (T * H * 2 + T * R * 4 + S * R * H * 2 / 8 + T * O * 4 + T * R * O / O * 2) / 1e9

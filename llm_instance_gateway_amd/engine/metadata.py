@@ -42,6 +42,11 @@ class BatchMeta:
 
     # Multi-LoRA (SGMV): per-token adapter slot index, -1 = base model only.
     lora_slot_per_token: Optional[torch.Tensor] = None  # [T] int32
+    # Prefill LoRA segment tiles (64-token runs sharing one adapter slot) for
+    # the MFMA SGMV kernels: token start, length (<=64), adapter slot.
+    lora_tile_t0: Optional[torch.Tensor] = None    # [ntiles] int32
+    lora_tile_len: Optional[torch.Tensor] = None   # [ntiles] int32
+    lora_tile_slot: Optional[torch.Tensor] = None  # [ntiles] int32
 
     @property
     def num_tokens(self) -> int:

@@ -223,13 +223,35 @@ class ModelRunner:
         mgr = getattr(self, "adapter_manager", None)
         if mgr is not None and any(ss.seq.lora_name for ss in prefills + decodes):
             slots: List[int] = []
+            # prefill tokens are sequence-contiguous: emit 64-token segment
+            # tiles per same-slot run for the MFMA SGMV path
+            t0s: List[int] = []
+            tlens: List[int] = []
+            tslots: List[int] = []
             for ss in prefills:
-                slots.extend([mgr.slot_of(ss.seq.lora_name)] * ss.num_tokens)
+                s_id = mgr.slot_of(ss.seq.lora_name)
+                base = len(slots)
+                slots.extend([s_id] * ss.num_tokens)
+                if s_id >= 0:
+                    for off in range(0, ss.num_tokens, 64):
+                        t0s.append(base + off)
+                        tlens.append(min(64, ss.num_tokens - off))
+                        tslots.append(s_id)
             for ss in decodes:
                 slots.append(mgr.slot_of(ss.seq.lora_name))
             meta.lora_slot_per_token = torch.tensor(slots, dtype=torch.int32).to(
                 dev, non_blocking=True
             )
+            if t0s:
+                meta.lora_tile_t0 = torch.tensor(t0s, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
+                meta.lora_tile_len = torch.tensor(tlens, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
+                meta.lora_tile_slot = torch.tensor(tslots, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
 
         ids_t = torch.tensor(input_ids, dtype=torch.int64).to(dev, non_blocking=True)
         pos_t = torch.tensor(positions, dtype=torch.int64).to(dev, non_blocking=True)

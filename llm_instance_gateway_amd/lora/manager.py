@@ -278,13 +278,24 @@ class AdapterManager:
                     max(T, 4096), self.rank, dtype=self.dtype, device=self.device
                 )
                 self._u_ws = ws
-            # wide kernels only for prefill batches (sequence-contiguous
-            # adapter segments); decode batches keep the per-token kernels
-            wide = bool(getattr(meta, "num_prefill_tokens", 0) > 0 and x.shape[0] >= 256)
-            ops._EXT.sgmv_apply(
-                y, x, self.a_stacks[key], self.b_stacks[key], ws, slots, 1.0,
-                wide, ops._stream(),
-            )
+            tile_t0 = getattr(meta, "lora_tile_t0", None)
+            if tile_t0 is not None and y.shape[1] % 64 == 0 and x.shape[1] % 32 == 0:
+                # prefill: MFMA over 64-token same-adapter segment tiles
+                ops._EXT.sgmv_apply_tiled(
+                    y, x, self.a_stacks[key], self.b_stacks[key], ws,
+                    tile_t0, meta.lora_tile_len, meta.lora_tile_slot, 1.0,
+                    ops._stream(),
+                )
+            else:
+                # decode (or odd dims): per-token kernels; wide variants only
+                # help when consecutive tokens share an adapter
+                wide = bool(
+                    getattr(meta, "num_prefill_tokens", 0) > 0 and x.shape[0] >= 256
+                )
+                ops._EXT.sgmv_apply(
+                    y, x, self.a_stacks[key], self.b_stacks[key], ws, slots,
+                    1.0, wide, ops._stream(),
+                )
         else:
             u = ops.sgmv_shrink(x, self.a_stacks[key], slots)
             ops.sgmv_expand(u, self.b_stacks[key], slots, y, 1.0)

@@ -35,6 +35,12 @@ void lig_sgmv_expand_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             hipStream_t);
 void lig_skinny_gemm_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             int, int, int, hipStream_t);
+void lig_sgmv_shrink_tiled_launch(uint16_t*, const uint16_t*, const uint16_t*,
+                                  const int*, const int*, const int*, int,
+                                  int, int, int, hipStream_t);
+void lig_sgmv_expand_tiled_launch(uint16_t*, const uint16_t*, const uint16_t*,
+                                  const int*, const int*, const int*, int,
+                                  float, int, int, int, hipStream_t);
 void lig_paged_attention_prefill_launch(
     uint16_t*, const uint16_t*, const uint16_t*, const uint16_t*, const int*,
     const int*, const int*, const int*, const int*, float, int, int, int,
@@ -204,6 +210,35 @@ void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
                          use_wide, st);
 }
 
+// Prefill LoRA: one shrink+expand over host-built 64-token segment tiles
+// (MFMA path; engine/model_runner.py builds the tiles).
+void sgmv_apply_tiled(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
+                      torch::Tensor b_stack, torch::Tensor u_ws,
+                      torch::Tensor tile_t0, torch::Tensor tile_len,
+                      torch::Tensor tile_slot, double scaling,
+                      int64_t stream) {
+  const int T = x.size(0);
+  const int H = x.size(1);
+  const int R = a_stack.size(1);
+  const int O = y.size(1);
+  const int ntiles = tile_t0.size(0);
+  TORCH_CHECK(tile_t0.scalar_type() == torch::kInt32);
+  TORCH_CHECK(u_ws.size(0) >= T && u_ws.size(1) == R, "u workspace too small");
+  TORCH_CHECK(R % 8 == 0 && R <= 64, "rank must be 8/16/32/64");
+  TORCH_CHECK(H % 32 == 0 && O % 64 == 0, "tiled sgmv needs H%32==0, O%64==0");
+  if (T == 0 || ntiles == 0) return;
+  hipStream_t st = as_stream(stream);
+  lig_sgmv_shrink_tiled_launch(bf16_ptr(u_ws), bf16_cptr(x),
+                               bf16_cptr(a_stack), tile_t0.data_ptr<int>(),
+                               tile_len.data_ptr<int>(),
+                               tile_slot.data_ptr<int>(), ntiles, T, R, H, st);
+  lig_sgmv_expand_tiled_launch(bf16_ptr(y), bf16_cptr(u_ws),
+                               bf16_cptr(b_stack), tile_t0.data_ptr<int>(),
+                               tile_len.data_ptr<int>(),
+                               tile_slot.data_ptr<int>(), ntiles,
+                               (float)scaling, T, R, O, st);
+}
+
 void skinny_gemm(torch::Tensor c, torch::Tensor a, torch::Tensor w,
                  int64_t stream) {
   const int M = a.size(0);
@@ -246,6 +281,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgmv_shrink", &sgmv_shrink);
   m.def("sgmv_apply", &sgmv_apply);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("sgmv_apply_tiled", &sgmv_apply_tiled);
   m.def("sgmv_expand", &sgmv_expand);
   m.def("mfma_probe", [](torch::Tensor out, torch::Tensor a, torch::Tensor b,
                          int64_t stream) {

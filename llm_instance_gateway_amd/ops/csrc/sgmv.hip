@@ -222,6 +222,148 @@ __global__ void sgmv_expand_wide_kernel(
   }
 }
 
+// ---- MFMA segment-tiled kernels (prefill) ---------------------------------
+//
+// Prefill batches carry 64-token segment tiles (host metadata: same adapter
+// slot per tile — engine/model_runner.py).  With the segment known per tile,
+// both halves become small MFMA GEMMs that read the adapter matrices ONCE
+// per tile instead of per token:
+//   shrink: per tile, u[64,R] = x[64,H] @ A_s[R,H]^T — K-loop over H.
+//   expand: per (tile, 64-out chunk), y[64,64] += u[64,R] @ B_s[64,R]^T —
+//           K = R zero-padded to the mfma's K=32.
+// Fragment layout as verified by csrc/mfma_probe.hip (see skinny_gemm.hip):
+// mfma_f32_16x16x32_bf16(a,b,c): C[i][j] += sum_k A[i,k]B[j,k]; a: lane li=i;
+// b: lane li=j; both k = lg*8..+7; C: row i = lg*4+r, col j = li.
+
+typedef __bf16 sgmv_bf16x8 __attribute__((ext_vector_type(8)));
+typedef float sgmv_f32x4 __attribute__((ext_vector_type(4)));
+
+template <int R>
+__global__ __launch_bounds__(256) void sgmv_shrink_tiled_kernel(
+    uint16_t* __restrict__ u,        // [T, R]
+    const uint16_t* __restrict__ x,  // [T, H]
+    const uint16_t* __restrict__ a,  // [S, R, H]
+    const int* __restrict__ tile_t0, const int* __restrict__ tile_len,
+    const int* __restrict__ tile_slot, const int T, const int H) {
+  constexpr int NR = (R + 15) / 16;  // 16-rank chunks
+  const int tile = blockIdx.x;
+  const int t0 = tile_t0[tile];
+  const int len = tile_len[tile];
+  const int s = tile_slot[tile];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int li = lane & 15;
+  const int lg = lane >> 4;
+
+  const int trow = wave * 16 + li;  // token within tile (a-fragment row)
+  const uint16_t* xrow = x + (size_t)(t0 + min(trow, len - 1)) * H;
+
+  sgmv_f32x4 acc[NR];
+#pragma unroll
+  for (int nr = 0; nr < NR; ++nr) acc[nr] = sgmv_f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const uint16_t* arow[NR];
+#pragma unroll
+  for (int nr = 0; nr < NR; ++nr) {
+    const int r = min(nr * 16 + li, R - 1);  // clamp for R < 16
+    arow[nr] = a + ((size_t)s * R + r) * H;
+  }
+
+#pragma unroll 4
+  for (int k0 = 0; k0 < H; k0 += 32) {
+    u16x8 araw = *reinterpret_cast<const u16x8*>(xrow + k0 + lg * 8);
+    sgmv_bf16x8 a_frag = *reinterpret_cast<sgmv_bf16x8*>(&araw);
+#pragma unroll
+    for (int nr = 0; nr < NR; ++nr) {
+      u16x8 braw = *reinterpret_cast<const u16x8*>(arow[nr] + k0 + lg * 8);
+      sgmv_bf16x8 b_frag = *reinterpret_cast<sgmv_bf16x8*>(&braw);
+      acc[nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
+                                                        acc[nr], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int r4 = 0; r4 < 4; ++r4) {
+    const int row = wave * 16 + lg * 4 + r4;  // C row = token within tile
+    if (row >= len) continue;
+#pragma unroll
+    for (int nr = 0; nr < NR; ++nr) {
+      const int rk = nr * 16 + li;
+      if (rk < R) u[(size_t)(t0 + row) * R + rk] = f32_to_bf16(acc[nr][r4]);
+    }
+  }
+}
+
+template <int R>
+__global__ __launch_bounds__(256) void sgmv_expand_tiled_kernel(
+    uint16_t* __restrict__ y,        // [T, O], accumulated
+    const uint16_t* __restrict__ u,  // [T, R]
+    const uint16_t* __restrict__ b,  // [S, O, R]
+    const int* __restrict__ tile_t0, const int* __restrict__ tile_len,
+    const int* __restrict__ tile_slot, const float scaling, const int T,
+    const int O) {
+  constexpr int KC = (R + 31) / 32;  // mfma K=32 chunks over the rank dim
+  const int o0 = blockIdx.x * 64;
+  const int tile = blockIdx.y;
+  const int t0 = tile_t0[tile];
+  const int len = tile_len[tile];
+  const int s = tile_slot[tile];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int li = lane & 15;
+  const int lg = lane >> 4;
+
+  // a fragments: u rows (token = t0 + wave*16 + li), zero-padded past R
+  const int trow = wave * 16 + li;
+  const uint16_t* urow = u + (size_t)(t0 + min(trow, len - 1)) * R;
+  sgmv_bf16x8 a_frag[KC];
+#pragma unroll
+  for (int kc = 0; kc < KC; ++kc) {
+    const int k = kc * 32 + lg * 8;
+    u16x8 raw;
+    if (k < R) {
+      raw = *reinterpret_cast<const u16x8*>(urow + k);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) raw[e] = 0;
+    }
+    a_frag[kc] = *reinterpret_cast<sgmv_bf16x8*>(&raw);
+  }
+
+  sgmv_f32x4 acc[4];  // 4 out chunks of 16
+#pragma unroll
+  for (int oc = 0; oc < 4; ++oc) {
+    acc[oc] = sgmv_f32x4{0.f, 0.f, 0.f, 0.f};
+    const uint16_t* brow = b + ((size_t)s * O + o0 + oc * 16 + li) * R;
+#pragma unroll
+    for (int kc = 0; kc < KC; ++kc) {
+      const int k = kc * 32 + lg * 8;
+      u16x8 raw;
+      if (k < R) {
+        raw = *reinterpret_cast<const u16x8*>(brow + k);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) raw[e] = 0;
+      }
+      sgmv_bf16x8 b_frag = *reinterpret_cast<sgmv_bf16x8*>(&raw);
+      acc[oc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[kc], b_frag,
+                                                        acc[oc], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int r4 = 0; r4 < 4; ++r4) {
+    const int row = wave * 16 + lg * 4 + r4;
+    if (row >= len) continue;
+    uint16_t* yp = y + (size_t)(t0 + row) * O + o0;
+#pragma unroll
+    for (int oc = 0; oc < 4; ++oc) {
+      const int col = oc * 16 + li;
+      yp[col] = f32_to_bf16(bf16_to_f32(yp[col]) + scaling * acc[oc][r4]);
+    }
+  }
+}
+
 // v2 kernels above stay the decode path (mixed per-token slots: more waves
 // per unit work, no serial reload chains); the _wide variants win only on
 // PREFILL batches, whose tokens are sequence-contiguous so a token group
@@ -276,6 +418,44 @@ void lig_sgmv_expand_launch(uint16_t* y, const uint16_t* u, const uint16_t* b,
     default: break;  // caller validates R % 8 == 0 && R <= 64
   }
 #undef LAUNCH_E
+}
+
+void lig_sgmv_shrink_tiled_launch(uint16_t* u, const uint16_t* x,
+                                  const uint16_t* a, const int* tile_t0,
+                                  const int* tile_len, const int* tile_slot,
+                                  int ntiles, int T, int R, int H,
+                                  hipStream_t stream) {
+#define LAUNCH_ST(RR)                                                       \
+  hipLaunchKernelGGL((sgmv_shrink_tiled_kernel<RR>), dim3(ntiles),          \
+                     dim3(256), 0, stream, u, x, a, tile_t0, tile_len,      \
+                     tile_slot, T, H)
+  switch (R) {
+    case 8: LAUNCH_ST(8); break;
+    case 16: LAUNCH_ST(16); break;
+    case 32: LAUNCH_ST(32); break;
+    case 64: LAUNCH_ST(64); break;
+    default: break;
+  }
+#undef LAUNCH_ST
+}
+
+void lig_sgmv_expand_tiled_launch(uint16_t* y, const uint16_t* u,
+                                  const uint16_t* b, const int* tile_t0,
+                                  const int* tile_len, const int* tile_slot,
+                                  int ntiles, float scaling, int T, int R,
+                                  int O, hipStream_t stream) {
+#define LAUNCH_ET(RR)                                                       \
+  hipLaunchKernelGGL((sgmv_expand_tiled_kernel<RR>), dim3(O / 64, ntiles),  \
+                     dim3(256), 0, stream, y, u, b, tile_t0, tile_len,      \
+                     tile_slot, scaling, T, O)
+  switch (R) {
+    case 8: LAUNCH_ET(8); break;
+    case 16: LAUNCH_ET(16); break;
+    case 32: LAUNCH_ET(32); break;
+    case 64: LAUNCH_ET(64); break;
+    default: break;
+  }
+#undef LAUNCH_ET
 }
 
 }  // extern "C"

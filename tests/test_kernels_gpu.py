@@ -227,3 +227,41 @@ def test_lora_engine_gpu_matches_cpu():
     gpu.generate_all()
     agree = sum(int(x == y) for x, y in zip(a.output_token_ids, b.output_token_ids))
     assert agree >= len(a.output_token_ids) * 0.65
+
+
+@pytest.mark.parametrize("R,O", [(16, 6144), (8, 128), (32, 4096)])
+def test_sgmv_tiled_matches_ref(R, O):
+    """MFMA segment-tiled prefill path (64-token same-slot tiles) vs the
+    per-token fp32 reference."""
+    torch.manual_seed(5)
+    H, S = 4096 if O > 256 else 64, 4
+    # synthetic prefill batch: runs of seq-contiguous tokens, some without lora
+    seq_lens = [70, 3, 64, 129, 40]
+    seq_slots = [0, -1, 2, 3, 1]
+    slots_l, t0s, lens, tslots = [], [], [], []
+    for n, s in zip(seq_lens, seq_slots):
+        base = len(slots_l)
+        slots_l += [s] * n
+        if s >= 0:
+            for off in range(0, n, 64):
+                t0s.append(base + off)
+                lens.append(min(64, n - off))
+                tslots.append(s)
+    T = len(slots_l)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV) * 0.1
+    a = torch.randn(S, R, H, dtype=torch.bfloat16, device=DEV) * 0.05
+    b = torch.randn(S, O, R, dtype=torch.bfloat16, device=DEV) * 0.05
+    slots = torch.tensor(slots_l, dtype=torch.int32, device=DEV)
+    y = torch.randn(T, O, dtype=torch.bfloat16, device=DEV) * 0.1
+    y_ref = y.clone().float()
+    u_ws = torch.zeros(T, R, dtype=torch.bfloat16, device=DEV)
+    ops._EXT.sgmv_apply_tiled(
+        y, x, a, b, u_ws,
+        torch.tensor(t0s, dtype=torch.int32, device=DEV),
+        torch.tensor(lens, dtype=torch.int32, device=DEV),
+        torch.tensor(tslots, dtype=torch.int32, device=DEV),
+        1.0, torch.cuda.current_stream().cuda_stream,
+    )
+    u_ref = torch_ref.sgmv_shrink(x.float(), a.float(), slots)
+    torch_ref.sgmv_expand(u_ref, b.float(), slots, y_ref, 1.0)
+    close(y, y_ref.to(torch.bfloat16))
