@@ -21,6 +21,7 @@ needs no per-token scaling factor.
 from __future__ import annotations
 
 import logging
+import os
 import threading
 import time
 from dataclasses import dataclass, field
@@ -29,6 +30,9 @@ from typing import Dict, List, Optional, Tuple
 import torch
 
 log = logging.getLogger(__name__)
+
+# A/B switch for the MFMA tiled prefill-LoRA path (benchmarking)
+_NO_TILED_SGMV = os.environ.get("LIG_NO_TILED_SGMV", "0") == "1"
 
 # module name -> (in_features, out_features) resolver
 def _module_dims(mc, tp_size: int) -> Dict[str, Tuple[int, int]]:
@@ -279,7 +283,12 @@ class AdapterManager:
                 )
                 self._u_ws = ws
             tile_t0 = getattr(meta, "lora_tile_t0", None)
-            if tile_t0 is not None and y.shape[1] % 64 == 0 and x.shape[1] % 32 == 0:
+            if (
+                tile_t0 is not None
+                and y.shape[1] % 64 == 0
+                and x.shape[1] % 32 == 0
+                and not _NO_TILED_SGMV
+            ):
                 # prefill: MFMA over 64-token same-adapter segment tiles
                 ops._EXT.sgmv_apply_tiled(
                     y, x, self.a_stacks[key], self.b_stacks[key], ws,
