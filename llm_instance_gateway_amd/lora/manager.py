@@ -31,8 +31,13 @@ import torch
 
 log = logging.getLogger(__name__)
 
-# A/B switch for the MFMA tiled prefill-LoRA path (benchmarking)
-_NO_TILED_SGMV = os.environ.get("LIG_NO_TILED_SGMV", "0") == "1"
+# MFMA segment-tiled prefill-LoRA path: wins its microbench (2.5x at
+# T=2048/O=28672, benchmarks/bench_sgmv.py) but LOSES end-to-end in the
+# engine (same-box A/B: 26.9 vs 31.0 req/s goodput) — real prefill batches
+# have many partial 64-token tiles and the scalar y read-modify-write
+# epilogue costs more than the adapter-read amortization saves.  Opt-in for
+# further work; the wide per-token kernels are the default.
+_TILED_SGMV = os.environ.get("LIG_TILED_SGMV", "0") == "1"
 
 # module name -> (in_features, out_features) resolver
 def _module_dims(mc, tp_size: int) -> Dict[str, Tuple[int, int]]:
@@ -287,7 +292,7 @@ class AdapterManager:
                 tile_t0 is not None
                 and y.shape[1] % 64 == 0
                 and x.shape[1] % 32 == 0
-                and not _NO_TILED_SGMV
+                and _TILED_SGMV
             ):
                 # prefill: MFMA over 64-token same-adapter segment tiles
                 ops._EXT.sgmv_apply_tiled(
