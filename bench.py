@@ -69,6 +69,8 @@ def main() -> None:
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--concurrency", type=int, default=256)
     ap.add_argument("--device", default=None, help="override (e.g. cpu for dev)")
+    ap.add_argument("--max-batched-tokens", type=int, default=0,
+                    help="prefill chunk budget override (0 = default)")
     ap.add_argument(
         "--lora",
         action=argparse.BooleanOptionalAction,
@@ -106,7 +108,7 @@ def main() -> None:
         model=mc,
         num_kv_blocks=8192 if is_cuda else 256,
         max_num_seqs=256,
-        max_num_batched_tokens=4096 if is_cuda else 512,
+        max_num_batched_tokens=(args.max_batched_tokens or (4096 if is_cuda else 512)),
         max_model_len=2048 if is_cuda else 512,
         gpu_memory_utilization=0.90,
         max_loras=NUM_LORAS,
