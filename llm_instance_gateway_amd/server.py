@@ -69,6 +69,11 @@ def main(argv: Optional[List[str]] = None) -> int:
     ap.add_argument("--port", type=int, default=8081, help="HTTP front-door port (reference gateway.yaml:13-15)")
     ap.add_argument("--model", default="", help="base model preset override")
     ap.add_argument("--devices", default="", help="comma list, e.g. cuda:0,cuda:1 or cpu")
+    ap.add_argument(
+        "--remote-engines", default="",
+        help="comma list of host:port model servers — remote-router mode "
+        "(the reference's topology: scrape + schedule + forward, no local GPUs)",
+    )
     ap.add_argument("--v", type=int, default=1, help="log verbosity (klog analogue)")
     args = ap.parse_args(argv)
 
@@ -81,6 +86,28 @@ def main(argv: Optional[List[str]] = None) -> int:
     from .serving.frontdoor import build_app
 
     cfg = load_config(args.config) if args.config else None
+
+    if args.remote_engines or (cfg and cfg.pool.engine_addresses and not cfg.pool.gpu_ids):
+        from .serving.remote_router import build_remote_router_app
+
+        addrs = (
+            args.remote_engines.split(",")
+            if args.remote_engines
+            else cfg.pool.engine_addresses
+        )
+        app = build_remote_router_app(
+            addrs,
+            models=cfg.models if cfg else None,
+            refresh_metrics_interval_s=(
+                cfg.refresh_metrics_interval_s if cfg else 0.050
+            ),
+        )
+        import uvicorn
+
+        port = args.port if args.port else (cfg.port if cfg else 8081)
+        log.info("remote-router mode over %d engines", len(addrs))
+        uvicorn.run(app, host="0.0.0.0", port=port, log_level="warning")
+        return 0
     pool = build_pool(args, cfg)
     log.info("starting %d engine(s)...", len(pool.handles))
     pool.start()
