@@ -158,6 +158,7 @@ def test_temperature_sampling_deterministic_per_seed():
     outs = []
     for _ in range(2):
         seqmod._seq_counter = itertools.count(5000)  # align seq ids
+        torch.manual_seed(17)  # identical weights
         eng = make_engine()
         seq = eng.add_request(
             prompts(1, 16)[0],
