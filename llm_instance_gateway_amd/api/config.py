@@ -219,6 +219,38 @@ def reconcile_adapters(cfg: GatewayConfig, pool) -> Tuple[Set[str], Set[str], Se
     return loaded, unloaded, dupes
 
 
+def reconcile_adapters_remote(cfg: GatewayConfig, actuator) -> Tuple[Set[str], Set[str], Set[str]]:
+    """Sidecar reconcile against a REMOTE server (`sidecar.py:215-239`): diff
+    ensureExist − ensureNotExist against the server's GET /v1/models listing
+    (`sidecar.py:143-156`), actuate over its dynamic-LoRA HTTP API.
+    ``actuator`` is a `router.remote.RemoteAdapterActuator`."""
+    exist = {a.id: a for a in cfg.ensure_exist}
+    not_exist = {a.id for a in cfg.ensure_not_exist}
+    dupes = set(exist) & not_exist
+    for d in sorted(dupes):
+        log.warning(
+            "skipping adapter %r: present in both ensureExist and ensureNotExist", d
+        )
+    listed = set(actuator.list_models())
+    loaded: Set[str] = set()
+    unloaded: Set[str] = set()
+    for aid, spec in exist.items():
+        if aid in dupes or aid in listed:
+            continue
+        err = actuator.load(aid, spec.source, spec.base_model)
+        if err:
+            log.error("remote load %s failed: %s", aid, err)
+        else:
+            loaded.add(aid)
+    for aid in (not_exist - dupes) & listed:
+        err = actuator.unload(aid)
+        if err:
+            log.error("remote unload %s failed: %s", aid, err)
+        else:
+            unloaded.add(aid)
+    return loaded, unloaded, dupes
+
+
 def wait_healthy(pool, timeout_s: float, interval_s: float) -> bool:
     """Sidecar health gate (`sidecar.py:158-175`): poll until an engine is
     serving or timeout."""
@@ -240,13 +272,21 @@ class ConfigReconciler:
     def __init__(
         self,
         path: str,
-        pool,
+        pool=None,
         poll_interval_s: float = 1.0,
         health_timeout_s: float = DEFAULT_HEALTH_TIMEOUT_S,
         health_interval_s: float = DEFAULT_HEALTH_INTERVAL_S,
+        datastore=None,
+        remote_actuators=None,
     ) -> None:
+        """In-process mode: pass ``pool``.  Remote (sidecar) mode: pass
+        ``datastore`` + ``remote_actuators`` (one per model server)."""
         self.path = path
         self.pool = pool
+        self.datastore = datastore if datastore is not None else (
+            pool.datastore if pool is not None else None
+        )
+        self.remote_actuators = list(remote_actuators or [])
         self.poll_interval_s = poll_interval_s
         self.health_timeout_s = health_timeout_s
         self.health_interval_s = health_interval_s
@@ -263,11 +303,21 @@ class ConfigReconciler:
         except (OSError, ConfigError, yaml.YAMLError) as e:
             log.error("config %s rejected: %s", self.path, e)
             return False
-        if not wait_healthy(self.pool, self.health_timeout_s, self.health_interval_s):
-            log.error("no healthy engine within %.0fs; skipping reconcile", self.health_timeout_s)
-            return False
-        reconcile_models(cfg, self.pool.datastore)
-        reconcile_adapters(cfg, self.pool)
+        if self.pool is not None:
+            if not wait_healthy(self.pool, self.health_timeout_s, self.health_interval_s):
+                log.error("no healthy engine within %.0fs; skipping reconcile", self.health_timeout_s)
+                return False
+        else:
+            for act in self.remote_actuators:
+                if not act.wait_healthy(self.health_timeout_s, self.health_interval_s):
+                    log.error("remote engine %s unhealthy; skipping reconcile", act.base_url)
+                    return False
+        if self.datastore is not None:
+            reconcile_models(cfg, self.datastore)
+        if self.pool is not None:
+            reconcile_adapters(cfg, self.pool)
+        for act in self.remote_actuators:
+            reconcile_adapters_remote(cfg, act)
         self.last_config = cfg
         self.reconcile_count += 1
         return True

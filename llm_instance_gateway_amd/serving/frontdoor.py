@@ -212,7 +212,13 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
                 loop = asyncio.get_running_loop()
                 sent = 0
                 while True:
-                    tok = await loop.run_in_executor(None, ticket.tokens.get)
+                    try:
+                        tok = await loop.run_in_executor(None, ticket.tokens.get)
+                    except asyncio.CancelledError:
+                        # client disconnected mid-stream: stop decoding for a
+                        # dead connection (engine abort + block free)
+                        pool.abort(rid)
+                        raise
                     if tok is None:
                         break
                     sent += 1

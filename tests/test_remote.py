@@ -111,3 +111,35 @@ def test_remote_adapter_actuator(fake_server):
         "base_model_name": "",
     }) in _Handler.calls
     assert ("/v1/unload_lora_adapter", {"lora_name": "a2"}) in _Handler.calls
+
+
+def test_remote_config_reconciler(fake_server, tmp_path):
+    """Sidecar mode against a remote server: diff vs GET /v1/models, actuate
+    over HTTP, models into a local datastore."""
+    from llm_instance_gateway_amd.api.config import ConfigReconciler
+    from llm_instance_gateway_amd.router.remote import RemoteAdapterActuator
+
+    path = tmp_path / "remote.yaml"
+    path.write_text(
+        "pool: {name: remote-pool, engine_addresses: ['%s']}\n"
+        "models: [{model_name: remote-model}]\n"
+        "ensureExist: {models: [{id: a1}, {id: fresh-adapter}]}\n"
+        "ensureNotExist: {models: [{id: a2}]}\n" % fake_server
+    )
+    ds = Datastore()
+    before = len(_Handler.calls)
+    rec = ConfigReconciler(
+        str(path),
+        datastore=ds,
+        remote_actuators=[RemoteAdapterActuator(f"http://{fake_server}")],
+        health_timeout_s=5,
+        health_interval_s=0.1,
+    )
+    assert rec.reconcile_once()
+    assert ds.fetch_model("remote-model") is not None
+    calls = _Handler.calls[before:]
+    # a1 already listed -> skipped; fresh-adapter loaded; a2 unloaded
+    paths = [(p, d.get("lora_name")) for p, d in calls]
+    assert ("/v1/load_lora_adapter", "fresh-adapter") in paths
+    assert ("/v1/unload_lora_adapter", "a2") in paths
+    assert ("/v1/load_lora_adapter", "a1") not in paths
