@@ -54,13 +54,17 @@ class MoEBlock(nn.Module):
         h = cfg.hidden_size
         # router is replicated (tiny GEMM, identical on every rank)
         self.gate = _linear(h, cfg.num_experts, device, dtype)
-        # [E, 2*inter, h] and [E, h, inter], bounded random init
+        # Stored PRE-TRANSPOSED for bmm ([E, h, 2*inter] and [E, inter, h]):
+        # feeding bmm a transposed non-contiguous weight view memory-faults
+        # in hipBLASLt's strided-batched path at large batch (reproduced at
+        # [8,1073,4096]x[8,4096,28672]^T on MI355X; contiguous operands are
+        # also the faster layout).
         self.w13 = nn.Parameter(
-            torch.empty(cfg.num_experts, 2 * inter, h, device=device, dtype=dtype)
+            torch.empty(cfg.num_experts, h, 2 * inter, device=device, dtype=dtype)
             .uniform_(-0.02, 0.02)
         )
         self.w2 = nn.Parameter(
-            torch.empty(cfg.num_experts, h, inter, device=device, dtype=dtype)
+            torch.empty(cfg.num_experts, inter, h, device=device, dtype=dtype)
             .uniform_(-0.02, 0.02)
         )
 
@@ -106,8 +110,8 @@ class MoEBlock(nn.Module):
         gather_idx.index_copy_(0, dest, sorted_tokens)
         x_pad = torch.cat([x, x.new_zeros(1, h)], dim=0)
         xe = x_pad.index_select(0, gather_idx).view(E, cap, h)
-        act = ops.silu_and_mul(torch.bmm(xe, self.w13.transpose(1, 2)).view(E * cap, -1))
-        ye = torch.bmm(act.view(E, cap, -1), self.w2.transpose(1, 2)).view(E * cap, h)
+        act = ops.silu_and_mul(torch.bmm(xe, self.w13).view(E * cap, -1))
+        ye = torch.bmm(act.view(E, cap, -1), self.w2).view(E * cap, h)
         w_pad = torch.zeros(E * cap, 1, dtype=x.dtype, device=dev)
         w_pad.index_copy_(0, dest, sorted_w.unsqueeze(1))
         out = torch.zeros_like(x)

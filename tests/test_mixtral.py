@@ -38,9 +38,9 @@ def test_moe_block_matches_per_token_reference():
     for t in range(x.shape[0]):
         for j in range(mc.num_experts_per_tok):
             ei = int(e[t, j])
-            gu = x[t] @ blk.w13[ei].t()
+            gu = x[t] @ blk.w13[ei]
             gate, up = gu.chunk(2)
-            ref[t] += w[t, j] * ((torch.nn.functional.silu(gate) * up) @ blk.w2[ei].t())
+            ref[t] += w[t, j] * ((torch.nn.functional.silu(gate) * up) @ blk.w2[ei])
     assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
 
 
