@@ -37,7 +37,7 @@ from dataclasses import dataclass, field
 from typing import Any, Callable, Dict, List, Optional, Sequence
 
 from ..router.datastore import Datastore
-from ..router.scheduler import ResourceExhausted, Scheduler
+from ..router.scheduler import ResourceExhausted, Scheduler, SchedulingError
 from ..router.types import Engine, EngineMetrics, EngineState, LLMRequest
 
 log = logging.getLogger(__name__)
@@ -670,7 +670,17 @@ class EnginePool:
         ticket = RequestTicket(req.request_id, engine_name, req.stream)
         with self._tickets_lock:
             self._tickets[req.request_id] = ticket
-        h.submit(req)
+        try:
+            h.submit(req)
+        except (BrokenPipeError, OSError) as e:
+            # engine died between routing and dispatch: drop it from the
+            # routing set now (don't wait for the watchdog tick) and surface
+            # a retryable scheduling error
+            h.healthy = False
+            self.datastore.remove_engine(engine_name)
+            with self._tickets_lock:
+                self._tickets.pop(req.request_id, None)
+            raise SchedulingError(f"engine {engine_name} unavailable: {e}") from e
         return ticket
 
     def route_and_submit(self, lreq: LLMRequest, preq: PoolRequest) -> RequestTicket:
