@@ -160,3 +160,28 @@ class TestScheduler:
         sched = Scheduler(FixedProvider(engines))
         with pytest.raises(ResourceExhausted):
             sched.schedule(LLMRequest(model="m", critical=False))
+
+
+def test_custom_thresholds_change_shed_behavior():
+    """The reference's compile-time consts are config knobs here
+    (scheduler.go:15-24 'make configurable' TODO)."""
+    from llm_instance_gateway_amd.router.filters import DropRequest, build_default_tree
+    from llm_instance_gateway_amd.router.types import EngineMetrics, EngineState, Engine, LLMRequest
+
+    states = [
+        EngineState(
+            engine=Engine("e0", 0),
+            metrics=EngineMetrics(waiting_queue_size=3, kv_cache_usage_percent=0.5),
+        )
+    ]
+    req = LLMRequest(model="m", critical=False)
+    # default thresholds (queue<=5, kv<=0.8): accepted
+    assert build_default_tree().apply(req, states)
+    # tightened queue threshold: dropped
+    import pytest as _pytest
+
+    with _pytest.raises(DropRequest):
+        build_default_tree(queue_threshold_critical=2).apply(req, states)
+    # tightened KV threshold: dropped
+    with _pytest.raises(DropRequest):
+        build_default_tree(kv_cache_threshold=0.4).apply(req, states)
