@@ -17,7 +17,7 @@ tests can pass a starlette TestClient.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Iterator, List, Optional, Sequence, Union
+from typing import Any, Dict, List, Optional, Sequence, Union
 
 
 class GatewayError(RuntimeError):

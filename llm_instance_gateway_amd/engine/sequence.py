@@ -12,7 +12,7 @@ from __future__ import annotations
 import enum
 import itertools
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 _seq_counter = itertools.count()

@@ -28,7 +28,7 @@ import bisect
 import re
 import threading
 import time
-from typing import Dict, Iterable, List, Optional, Tuple
+from typing import Dict, Iterable, Optional, Tuple
 
 from ..router.types import EngineMetrics
 

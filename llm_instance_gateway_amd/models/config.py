@@ -10,7 +10,7 @@ initialised randomly and the benchmark is synthetic-data.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

@@ -18,7 +18,7 @@ Weights are random-init (no network — synthetic serving benchmark).
 from __future__ import annotations
 
 import math
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.nn as nn

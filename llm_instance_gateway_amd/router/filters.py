@@ -22,7 +22,7 @@ queueing_threshold_lora=50 — made configurable here (the reference carries
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Callable, List, Optional, Tuple
+from typing import Callable, List, Optional
 
 from .types import EngineState, LLMRequest
 

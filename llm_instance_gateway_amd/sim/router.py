@@ -14,9 +14,8 @@ from observation; in a sim both converge to the same signal).
 
 from __future__ import annotations
 
-import math
 import random as _random
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 from .core import (
     MAX_KV_PERC_BEFORE_RECOMPUTE,
