@@ -129,16 +129,16 @@ def test_eos_stops_generation():
     eng = make_engine()
     seq = eng.add_request(
         prompts(1, 24)[0],
-        SamplingParams(max_new_tokens=64, ignore_eos=False, eos_token_id=-2),
+        SamplingParams(max_new_tokens=16, ignore_eos=False, eos_token_id=-2),
     )
     # impossible eos id -2 -> runs to max; then a real eos check
     done = eng.generate_all()
-    assert done[0].num_output_tokens == 64
+    assert done[0].num_output_tokens == 16
 
     eng2 = make_engine()
     seq2 = eng2.add_request(
         prompts(1, 24, seed=3)[0],
-        SamplingParams(max_new_tokens=64, ignore_eos=False, eos_token_id=0),
+        SamplingParams(max_new_tokens=16, ignore_eos=False, eos_token_id=0),
     )
     eng2.generate_all()
     out = seq2.output_token_ids
