@@ -294,7 +294,14 @@ def _worker_main(
         draining = nonlocal_draining[0]
 
         if eng.has_work():
-            eng.step()
+            try:
+                eng.step()
+            except Exception as e:  # OOM / kernel error: report, then die so
+                # the watchdog fails in-flight tickets and drops the engine
+                import traceback
+
+                send(("error", None, f"engine step failed: {e}\n{traceback.format_exc()}"))
+                raise
             finished = eng.collect_finished()
             if is_front and token_buf:
                 conn.send(("tokens", token_buf))
