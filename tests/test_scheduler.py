@@ -185,3 +185,14 @@ def test_custom_thresholds_change_shed_behavior():
     # tightened KV threshold: dropped
     with _pytest.raises(DropRequest):
         build_default_tree(kv_cache_threshold=0.4).apply(req, states)
+
+
+def test_pool_ref_falls_back_when_no_match():
+    """A pool_ref with no matching engines must not strand the request —
+    fall back to the full candidate set (extension safety net)."""
+    states = three_engines_healthy()  # untagged engines
+    sched = Scheduler(FixedProvider(states), rng=random.Random(0))
+    eng = sched.schedule(
+        LLMRequest(model="m", critical=True, pool_ref="nonexistent-pool")
+    )
+    assert eng.name in {"e1", "e2", "e3"}
