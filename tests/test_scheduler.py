@@ -187,12 +187,20 @@ def test_custom_thresholds_change_shed_behavior():
         build_default_tree(kv_cache_threshold=0.4).apply(req, states)
 
 
-def test_pool_ref_falls_back_when_no_match():
-    """A pool_ref with no matching engines must not strand the request —
-    fall back to the full candidate set (extension safety net)."""
-    states = three_engines_healthy()  # untagged engines
-    sched = Scheduler(FixedProvider(states), rng=random.Random(0))
+def test_pool_ref_scoping_and_fallback():
+    """pool_ref scopes candidates to tagged + untagged engines; when NOTHING
+    matches, fall back to the full set rather than stranding the request."""
+    tagged = [
+        EngineState(engine=Engine("p1-a", pool="pool-1"), metrics=EngineMetrics()),
+        EngineState(engine=Engine("p2-a", pool="pool-2"), metrics=EngineMetrics()),
+    ]
+    sched = Scheduler(FixedProvider(tagged), rng=random.Random(0))
+    for _ in range(5):
+        assert sched.schedule(
+            LLMRequest(model="m", critical=True, pool_ref="pool-2")
+        ).name == "p2-a"
+    # no engine carries this pool tag -> full-set fallback
     eng = sched.schedule(
         LLMRequest(model="m", critical=True, pool_ref="nonexistent-pool")
     )
-    assert eng.name in {"e1", "e2", "e3"}
+    assert eng.name in {"p1-a", "p2-a"}
