@@ -12,7 +12,7 @@ from __future__ import annotations
 import enum
 import itertools
 import time
-from dataclasses import dataclass
+from dataclasses import dataclass, field
 from typing import List, Optional
 
 _seq_counter = itertools.count()
@@ -35,6 +35,9 @@ class SamplingParams:
     ignore_eos: bool = True  # synthetic workloads have no meaningful EOS
     eos_token_id: int = -1
     seed: int = 0
+    # OpenAI `stop`: token-id sequences; generation halts when the output
+    # tail matches one, and the matched tail is trimmed from the output.
+    stop_sequences: List[List[int]] = field(default_factory=list)
 
 
 class Sequence:
@@ -69,6 +72,7 @@ class Sequence:
         self.first_token_time: Optional[float] = None
         self.finish_time: Optional[float] = None
         self.num_preemptions = 0
+        self.finish_reason: str = "length"
 
     # -- lengths ------------------------------------------------------------
     @property
@@ -112,6 +116,7 @@ class Sequence:
     def check_stop(self) -> bool:
         s = self.sampling
         if self.num_output_tokens >= s.max_new_tokens:
+            self.finish_reason = "length"
             return True
         if (
             not s.ignore_eos
@@ -119,7 +124,15 @@ class Sequence:
             and self.output_token_ids
             and self.output_token_ids[-1] == s.eos_token_id
         ):
+            self.finish_reason = "stop"
             return True
+        for stop in s.stop_sequences:
+            n = len(stop)
+            if n and self.num_output_tokens >= n and self.output_token_ids[-n:] == stop:
+                # OpenAI semantics: the stop text itself is not returned
+                del self.output_token_ids[-n:]
+                self.finish_reason = "stop"
+                return True
         return False
 
     def preempt(self) -> None:

@@ -54,6 +54,7 @@ class CompletionRequest(BaseModel):
     temperature: float = 0.0
     top_p: float = 1.0
     stream: bool = False
+    stop: Optional[Union[str, List[str]]] = None
     # non-standard extensions (the bench/tests use these)
     lora: Optional[str] = None  # explicit adapter override
     request_id: Optional[str] = None
@@ -134,6 +135,9 @@ class Gateway:
             max_new_tokens=req.max_tokens,
             pool_ref=im.pool_ref,
         )
+        stops = req.stop if req.stop is not None else []
+        if isinstance(stops, str):
+            stops = [stops]
         preq = PoolRequest(
             request_id=req.request_id or f"cmpl-{uuid.uuid4().hex[:24]}",
             prompt_token_ids=prompt_ids,
@@ -143,6 +147,7 @@ class Gateway:
             lora_name=lora,
             critical=critical,
             stream=req.stream,
+            stop_sequences=[tokenizer.encode(t, add_bos=False) for t in stops if t],
         )
         self.spans.event(preq.request_id, "arrived", model=req.model,
                          target=target, critical=critical)

@@ -68,6 +68,7 @@ class PoolRequest:
     lora_name: str = ""
     critical: bool = False
     stream: bool = False
+    stop_sequences: List[List[int]] = field(default_factory=list)
     arrival_time: float = field(default_factory=time.monotonic)
 
 
@@ -226,6 +227,7 @@ def _worker_main(
                         max_new_tokens=r.max_new_tokens,
                         temperature=r.temperature,
                         top_p=r.top_p,
+                        stop_sequences=r.stop_sequences,
                     ),
                     lora_name=r.lora_name,
                     request_id=r.request_id,
@@ -321,6 +323,7 @@ def _worker_main(
                             ttft=seq.ttft(),
                             tpot=seq.tpot(),
                             e2e=(seq.finish_time or now) - seq.arrival_time,
+                            finish_reason=seq.finish_reason,
                         )
                     )
                 conn.send(("finished", out))

@@ -181,3 +181,27 @@ def test_abort_frees_kv_blocks():
     eng.step()
     assert eng.scheduler.kv_usage < used_before
     assert not eng.has_work()
+
+
+def test_stop_sequences_halt_and_trim():
+    """OpenAI `stop` semantics: halt when the output tail matches a stop
+    token sequence; the matched tail is not returned; finish_reason=stop."""
+    from llm_instance_gateway_amd.engine.sequence import SamplingParams
+
+    torch.manual_seed(2)
+    eng = make_engine()
+    # discover what the model greedily emits, then stop on tokens 3..4
+    probe = eng.add_request(prompts(1, 20)[0], SamplingParams(max_new_tokens=8))
+    eng.generate_all()
+    emitted = list(probe.output_token_ids)
+    stop = emitted[3:5]
+
+    torch.manual_seed(2)
+    eng2 = make_engine()
+    seq = eng2.add_request(
+        prompts(1, 20)[0],
+        SamplingParams(max_new_tokens=8, stop_sequences=[stop]),
+    )
+    eng2.generate_all()
+    assert seq.output_token_ids == emitted[:3]  # stop tail trimmed
+    assert seq.finish_reason == "stop"
