@@ -55,6 +55,8 @@ class CompletionRequest(BaseModel):
     top_p: float = 1.0
     stream: bool = False
     stop: Optional[Union[str, List[str]]] = None
+    n: int = 1            # number of choices (non-streaming only)
+    echo: bool = False    # prepend the prompt text to each choice
     # non-standard extensions (the bench/tests use these)
     lora: Optional[str] = None  # explicit adapter override
     request_id: Optional[str] = None
@@ -115,7 +117,8 @@ class Gateway:
         return im, target
 
     async def complete(
-        self, req: CompletionRequest, critical_override: Optional[bool] = None
+        self, req: CompletionRequest, critical_override: Optional[bool] = None,
+        choice_idx: int = 0,
     ) -> tuple:
         """Route + dispatch + await; returns (ticket|None already raised)."""
         t0 = time.monotonic()
@@ -138,8 +141,9 @@ class Gateway:
         stops = req.stop if req.stop is not None else []
         if isinstance(stops, str):
             stops = [stops]
+        base_id = req.request_id or f"cmpl-{uuid.uuid4().hex[:24]}"
         preq = PoolRequest(
-            request_id=req.request_id or f"cmpl-{uuid.uuid4().hex[:24]}",
+            request_id=base_id if choice_idx == 0 else f"{base_id}-c{choice_idx}",
             prompt_token_ids=prompt_ids,
             max_new_tokens=req.max_tokens,
             temperature=req.temperature,
@@ -261,30 +265,50 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
                 },
             )
 
+        # n > 1: route + dispatch the remaining choices concurrently
+        extra_tickets = []
+        for ci in range(1, max(1, min(req.n, 16))):
+            try:
+                t_extra, _ = await gw.complete(req, choice_idx=ci)
+                extra_tickets.append(t_extra)
+            except (ResourceExhausted, SchedulingError):
+                break  # partial degradation: return the choices we can serve
+        tickets = [ticket] + extra_tickets
         try:
-            await _await_ticket(ticket)
+            for t in tickets:
+                await _await_ticket(t)
         except TimeoutError:
-            pool.abort(rid)
+            for t in tickets:
+                pool.abort(t.request_id)
             return JSONResponse(status_code=504, content={"error": "request timed out"})
         except RuntimeError as e:
             gw.metrics.requests_total.inc(model=req.model, outcome="error")
             return JSONResponse(status_code=500, content={"error": str(e)})
-        gw.record_finished(ticket, req.model)
+        echo_text = (
+            req.prompt if req.echo and isinstance(req.prompt, str) else ""
+        )
+        choices = []
+        total_completion = 0
+        for i, t in enumerate(tickets):
+            gw.record_finished(t, req.model)
+            fin_i = t.result
+            total_completion += len(fin_i.output_token_ids)
+            choices.append(
+                {
+                    "index": i,
+                    "text": echo_text + tokenizer.decode(fin_i.output_token_ids),
+                    "token_ids": fin_i.output_token_ids,
+                    "finish_reason": fin_i.finish_reason,
+                }
+            )
         fin = ticket.result
         body = {
             "id": rid,
             "object": "text_completion",
             "created": created,
             "model": req.model,
-            "choices": [
-                {
-                    "index": 0,
-                    "text": tokenizer.decode(fin.output_token_ids),
-                    "token_ids": fin.output_token_ids,
-                    "finish_reason": fin.finish_reason,
-                }
-            ],
-            "usage": _usage(fin.prompt_len, len(fin.output_token_ids)),
+            "choices": choices,
+            "usage": _usage(fin.prompt_len, total_completion),
         }
         return JSONResponse(
             content=body,

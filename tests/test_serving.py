@@ -257,3 +257,18 @@ def test_engine_crash_detected_and_restarted():
         assert eng.name == "engine-0"
     finally:
         p.stop()
+
+
+def test_n_choices_and_echo(client):
+    r = client.post(
+        "/v1/completions",
+        json={"model": "my-model", "prompt": "multi", "max_tokens": 3,
+              "n": 3, "echo": True},
+    )
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert [c["index"] for c in body["choices"]] == [0, 1, 2]
+    for c in body["choices"]:
+        assert c["text"].startswith("multi")  # echo prepends the prompt
+        assert len(c["token_ids"]) == 3
+    assert body["usage"]["completion_tokens"] == 9
