@@ -125,16 +125,21 @@ def build_remote_router_app(
             gw_metrics.requests_total.inc(model=req.model, outcome="error")
             return JSONResponse(status_code=502, content={"error": str(e)})
         gw_metrics.requests_total.inc(model=req.model, outcome="ok")
-        payload = resp.json() if resp.headers.get("content-type", "").startswith("application/json") else resp.text
+        is_json = resp.headers.get("content-type", "").startswith("application/json")
+        gw_metrics.e2e.observe(time.monotonic() - t0)
+        fwd_headers = {TARGET_ENGINE_HEADER: engine.name, RESP_DEBUG_HEADER: "true"}
+        if not is_json:
+            # pass non-JSON upstream bodies through verbatim
+            return PlainTextResponse(
+                resp.text, status_code=resp.status_code, headers=fwd_headers
+            )
+        payload = resp.json()
         usage = payload.get("usage") if isinstance(payload, dict) else None
         if usage:
             gw_metrics.prompt_tokens_total.inc(usage.get("prompt_tokens", 0))
             gw_metrics.completion_tokens_total.inc(usage.get("completion_tokens", 0))
-        gw_metrics.e2e.observe(time.monotonic() - t0)
         return JSONResponse(
-            status_code=resp.status_code,
-            content=payload,
-            headers={TARGET_ENGINE_HEADER: engine.name, RESP_DEBUG_HEADER: "true"},
+            status_code=resp.status_code, content=payload, headers=fwd_headers
         )
 
     @app.get("/health")
