@@ -65,8 +65,24 @@ class EngineScheduler:
 
     # -- intake --------------------------------------------------------------
     def add_seq(self, seq: Sequence) -> None:
+        """FIFO within a criticality tier; Critical requests are admitted
+        ahead of waiting Default/Sheddable ones (the engine-level half of the
+        reference's priority tiers — the router half sheds, scheduler.go:74-90;
+        preempted sequences re-queue at the very front regardless, matching
+        the sim's recompute priority)."""
         seq.status = SequenceStatus.WAITING
-        self.waiting.append(seq)
+        if seq.critical:
+            # insert after the last waiting critical (stable within tier),
+            # before the first non-critical fresh arrival
+            idx = 0
+            for i, w in enumerate(self.waiting):
+                if w.critical or w.num_computed_tokens > 0 or w.num_preemptions > 0:
+                    idx = i + 1
+                else:
+                    break
+            self.waiting.insert(idx, seq)
+        else:
+            self.waiting.append(seq)
 
     def abort_seq(self, request_id: str) -> bool:
         for q in (list(self.waiting), self.running):

@@ -205,3 +205,20 @@ def test_stop_sequences_halt_and_trim():
     eng2.generate_all()
     assert seq.output_token_ids == emitted[:3]  # stop tail trimmed
     assert seq.finish_reason == "stop"
+
+
+def test_critical_requests_admitted_first():
+    """Engine-level priority: a Critical arrival overtakes waiting
+    non-critical requests (but not preempted/recompute work)."""
+    eng = make_engine(max_num_seqs=1)  # force queueing behind one slot
+    occupant = eng.add_request(prompts(1, 16)[0], SamplingParams(max_new_tokens=30))
+    eng.step()  # occupant running
+    normal = eng.add_request(prompts(1, 16, seed=1)[0], SamplingParams(max_new_tokens=2))
+    crit = eng.add_request(
+        prompts(1, 16, seed=2)[0], SamplingParams(max_new_tokens=2), critical=True
+    )
+    assert [s.request_id for s in eng.scheduler.waiting] == [
+        crit.request_id, normal.request_id
+    ]
+    eng.generate_all()
+    assert crit.first_scheduled_time < normal.first_scheduled_time
