@@ -490,10 +490,11 @@ class EngineHandle:
         self.conn.send(("drain", on))
 
     def stop(self) -> None:
-        try:
-            self.conn.send(("stop",))
-        except (BrokenPipeError, OSError):
-            pass
+        if self.conn is not None:
+            try:
+                self.conn.send(("stop",))
+            except (BrokenPipeError, OSError):
+                pass
         for p in self.procs:
             p.join(timeout=10)
             if p.is_alive():
