@@ -283,7 +283,8 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
             return JSONResponse(status_code=504, content={"error": "request timed out"})
         except RuntimeError as e:
             gw.metrics.requests_total.inc(model=req.model, outcome="error")
-            return JSONResponse(status_code=500, content={"error": str(e)})
+            status = 400 if "max_model_len" in str(e) else 500
+            return JSONResponse(status_code=status, content={"error": str(e)})
         echo_text = (
             req.prompt if req.echo and isinstance(req.prompt, str) else ""
         )

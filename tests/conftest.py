@@ -12,6 +12,9 @@ def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: tests that require an MI355X GPU (run via gpurun)"
     )
+    # hang protection for the driver's round checks (pytest-timeout)
+    if config.getoption("--timeout", None) in (None, 0):
+        config.option.timeout = 600
 
 
 def pytest_collection_modifyitems(config, items):

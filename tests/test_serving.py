@@ -272,3 +272,12 @@ def test_n_choices_and_echo(client):
         assert c["text"].startswith("multi")  # echo prepends the prompt
         assert len(c["token_ids"]) == 3
     assert body["usage"]["completion_tokens"] == 9
+
+
+def test_overlong_prompt_rejected_400(client):
+    r = client.post(
+        "/v1/completions",
+        json={"model": "my-model", "prompt": list(range(1, 400)), "max_tokens": 2},
+    )
+    assert r.status_code == 400
+    assert "max_model_len" in r.json()["error"]
