@@ -92,15 +92,37 @@ class SimRouter:
             affine = [e for e in self.engines if req.lora in e.lora_loaded]
             if affine:
                 cands = affine
-        under_slo = [
-            e
-            for e in cands
-            if not e.is_saturated()
-            and self.estimate_latency(e, req)
-            <= req.target_latency_per_token * req.output_size
-            + prefill_delay(req.input_size)
-        ]
+        # TPOT-SLO gate: would this engine's per-token decode latency (with
+        # this request on board) still meet the request's target?  This is
+        # the signal that actually segregates the tiers — tight-SLO traffic
+        # stays off heavily-batched engines, loose-SLO traffic gets PACKED
+        # onto the busiest qualifying engine, keeping lightly-loaded engines
+        # free for the tight tier.
+        def qualifying(pool):
+            return [
+                e
+                for e in pool
+                if not e.is_saturated()
+                and decode_delay(
+                    e.tokens_in_decode() + req.input_size + req.output_size,
+                    len(e.decode) + 1,
+                )
+                <= req.target_latency_per_token
+            ]
+
+        under_slo = qualifying(cands)
+        if not under_slo and cands is not self.engines:
+            # adapter affinity must not trap a tight-SLO request on a packed
+            # engine: loading the adapter elsewhere beats missing the SLO
+            under_slo = qualifying(self.engines)
         if under_slo:
+            # pack the latency-TOLERANT tier onto the busiest qualifying
+            # engine (consolidation); SPREAD the tight tier onto the least
+            # loaded one — max-pending packing of tight-SLO traffic
+            # overshoots by the time the request reaches decode
+            tight = req.target_latency_per_token <= 0.05
+            if tight:
+                return min(under_slo, key=lambda e: len(e.decode) + e.queue_len())
             return max(under_slo, key=lambda e: e.pending_tokens())
         unsat = [e for e in cands if not e.is_saturated()]
         if unsat:
