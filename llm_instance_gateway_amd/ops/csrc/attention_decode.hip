@@ -27,7 +27,11 @@
 
 #include "common.h"
 
-template <int HEAD_DIM>
+// SPLIT=true: flash-decoding — blockIdx.z selects a context slice; the
+// un-normalized partial (oacc, m, l) goes to scratch fp32
+// [N, Hq, nsplits, D+2] and a combine kernel merges the slices.  Doubles
+// wave-level parallelism for long contexts (ROADMAP #1).
+template <int HEAD_DIM, bool SPLIT = false>
 __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
     uint16_t* __restrict__ out,            // [N, Hq, D]
     const uint16_t* __restrict__ q,        // [N, Hq, D]
@@ -39,7 +43,9 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
     const int num_q_heads,
     const int num_kv_heads,
     const int block_size,
-    const int max_blocks) {
+    const int max_blocks,
+    float* __restrict__ scratch = nullptr,  // [N, Hq, nsplits, D+2]
+    const int nsplits = 1) {
   constexpr int DL = HEAD_DIM / 8;  // dims per lane
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
@@ -79,6 +85,12 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
 
   const int* bt = block_tables + (size_t)seq * max_blocks;
   const int iters = (ctx + 7) / 8;
+  int it0 = 0, it1 = iters;
+  if constexpr (SPLIT) {
+    const int span = (iters + nsplits - 1) / nsplits;
+    it0 = blockIdx.z * span;
+    it1 = min(iters, it0 + span);
+  }
 
   // Software-pipelined K/V: the next iteration's vector loads are issued
   // BEFORE this iteration's dot/softmax/accumulate chain (whose xor-shuffle
@@ -160,12 +172,12 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
         oacc[i] = oacc[i] * alpha + p * vf[i];                               \
   } while (0)
 
-  if (iters > 0) ISSUE_LOAD(kraw0, vraw0, valid0, 0);
-  for (int it = 0; it < iters; it += 2) {
-    if (it + 1 < iters) ISSUE_LOAD(kraw1, vraw1, valid1, it + 1);
+  if (it1 > it0) ISSUE_LOAD(kraw0, vraw0, valid0, it0);
+  for (int it = it0; it < it1; it += 2) {
+    if (it + 1 < it1) ISSUE_LOAD(kraw1, vraw1, valid1, it + 1);
     STEP_BODY(kraw0, vraw0, valid0);
-    if (it + 1 >= iters) break;
-    if (it + 2 < iters) ISSUE_LOAD(kraw0, vraw0, valid0, it + 2);
+    if (it + 1 >= it1) break;
+    if (it + 2 < it1) ISSUE_LOAD(kraw0, vraw0, valid0, it + 2);
     STEP_BODY(kraw1, vraw1, valid1);
   }
 #undef ISSUE_LOAD
@@ -174,6 +186,23 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
   // combine the 8 token-subsets (lanes stride-8 apart share dchunk)
 #pragma unroll
   for (int i = 0; i < DL; ++i) oacc[i] = stride_reduce_sum<8, 8>(oacc[i]);
+
+  if constexpr (SPLIT) {
+    // un-normalized partial out; m and l are wave-uniform after the
+    // stride reduces above
+    if (tok == 0) {
+      float* sp = scratch +
+          (((size_t)seq * num_q_heads + qh) * nsplits + blockIdx.z) *
+              (HEAD_DIM + 2);
+#pragma unroll
+      for (int i = 0; i < DL; ++i) sp[dchunk * DL + i] = oacc[i];
+      if (dchunk == 0) {
+        sp[HEAD_DIM] = m;
+        sp[HEAD_DIM + 1] = l;
+      }
+    }
+    return;
+  }
 
   if (tok == 0) {
     const float inv_l = (l > 0.0f) ? 1.0f / l : 0.0f;
@@ -192,6 +221,76 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
       for (int i = 0; i < DL; ++i) op[i] = f32_to_bf16(oacc[i] * inv_l);
     }
   }
+}
+
+template <int HEAD_DIM>
+__global__ void paged_attention_decode_combine_kernel(
+    uint16_t* __restrict__ out,           // [N, Hq, D]
+    const float* __restrict__ scratch,    // [N, Hq, nsplits, D+2]
+    const int num_q_heads, const int nsplits) {
+  // one wave per (seq, q-head); lane covers D/64 dims
+  const int seq = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int lane = threadIdx.x;
+  constexpr int DPL = HEAD_DIM >= 64 ? HEAD_DIM / 64 : 1;
+  const float* base =
+      scratch + ((size_t)seq * num_q_heads + qh) * (size_t)nsplits *
+                    (HEAD_DIM + 2);
+  float gm = -INFINITY;
+  for (int sI = 0; sI < nsplits; ++sI)
+    gm = fmaxf(gm, base[(size_t)sI * (HEAD_DIM + 2) + HEAD_DIM]);
+  float gl = 0.0f;
+  float acc[DPL];
+#pragma unroll
+  for (int i = 0; i < DPL; ++i) acc[i] = 0.0f;
+  for (int sI = 0; sI < nsplits; ++sI) {
+    const float* sp = base + (size_t)sI * (HEAD_DIM + 2);
+    const float ms = sp[HEAD_DIM];
+    if (ms == -INFINITY) continue;
+    const float w = __expf(ms - gm);
+    gl += sp[HEAD_DIM + 1] * w;
+#pragma unroll
+    for (int i = 0; i < DPL; ++i) {
+      const int d = lane * DPL + i;
+      if (d < HEAD_DIM) acc[i] += sp[d] * w;
+    }
+  }
+  const float inv = gl > 0.0f ? 1.0f / gl : 0.0f;
+  uint16_t* op = out + ((size_t)seq * num_q_heads + qh) * HEAD_DIM;
+#pragma unroll
+  for (int i = 0; i < DPL; ++i) {
+    const int d = lane * DPL + i;
+    if (d < HEAD_DIM) op[d] = f32_to_bf16(acc[i] * inv);
+  }
+}
+
+extern "C" void lig_paged_attention_decode_split_launch(
+    uint16_t* out, const uint16_t* q, const uint16_t* k_cache,
+    const uint16_t* v_cache, const int* block_tables,
+    const int* context_lens, float* scratch, float scale, int num_seqs,
+    int num_q_heads, int num_kv_heads, int head_dim, int block_size,
+    int max_blocks, int nsplits, hipStream_t stream) {
+  const int group = num_q_heads / num_kv_heads;
+  dim3 grid(num_seqs, num_kv_heads, nsplits);
+  dim3 block(group * WAVE_SIZE);
+#define LAUNCH_S(HD)                                                        \
+  do {                                                                      \
+    hipLaunchKernelGGL((paged_attention_decode_kernel<HD, true>), grid,     \
+                       block, 0, stream, out, q, k_cache, v_cache,          \
+                       block_tables, context_lens, scale, num_q_heads,      \
+                       num_kv_heads, block_size, max_blocks, scratch,       \
+                       nsplits);                                            \
+    hipLaunchKernelGGL((paged_attention_decode_combine_kernel<HD>),         \
+                       dim3(num_seqs, num_q_heads), dim3(WAVE_SIZE), 0,     \
+                       stream, out, scratch, num_q_heads, nsplits);         \
+  } while (0)
+  switch (head_dim) {
+    case 16: LAUNCH_S(16); break;
+    case 64: LAUNCH_S(64); break;
+    case 128: LAUNCH_S(128); break;
+    default: break;
+  }
+#undef LAUNCH_S
 }
 
 extern "C" void lig_paged_attention_decode_launch(

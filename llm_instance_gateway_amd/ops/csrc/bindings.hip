@@ -26,6 +26,10 @@ void lig_paged_attention_decode_launch(uint16_t*, const uint16_t*,
                                        const uint16_t*, const uint16_t*,
                                        const int*, const int*, float, int,
                                        int, int, int, int, int, hipStream_t);
+void lig_paged_attention_decode_split_launch(
+    uint16_t*, const uint16_t*, const uint16_t*, const uint16_t*,
+    const int*, const int*, float*, float, int, int, int, int, int, int,
+    int, hipStream_t);
 void lig_mfma_probe_launch(float*, const uint16_t*, const uint16_t*,
                            hipStream_t);
 void lig_sgmv_shrink_launch(uint16_t*, const uint16_t*, const uint16_t*,
@@ -144,6 +148,37 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
       bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
       block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
       (float)scale, N, Hq, Hkv, D, BS, max_blocks, as_stream(stream));
+}
+
+// Flash-decoding variant: context split across nsplits workgroups with an
+// fp32 partial scratch [N, Hq, nsplits, D+2] and a combine pass (ROADMAP #1;
+// experimental — not yet on the default dispatch path).
+void paged_attention_decode_split(torch::Tensor out, torch::Tensor q,
+                                  torch::Tensor k_cache,
+                                  torch::Tensor v_cache,
+                                  torch::Tensor block_tables,
+                                  torch::Tensor context_lens,
+                                  torch::Tensor scratch, double scale,
+                                  int64_t nsplits, int64_t stream) {
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  TORCH_CHECK(context_lens.scalar_type() == torch::kInt32);
+  TORCH_CHECK(scratch.scalar_type() == torch::kFloat32);
+  const int N = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(D == 16 || D == 64 || D == 128, "head_dim must be 16/64/128");
+  TORCH_CHECK(Hq % Hkv == 0 && Hq / Hkv <= 16, "bad GQA group size");
+  TORCH_CHECK(scratch.numel() >= (int64_t)N * Hq * nsplits * (D + 2),
+              "scratch too small");
+  if (N == 0) return;
+  lig_paged_attention_decode_split_launch(
+      bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
+      block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
+      scratch.data_ptr<float>(), (float)scale, N, Hq, Hkv, D, BS,
+      max_blocks, (int)nsplits, as_stream(stream));
 }
 
 void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
@@ -277,6 +312,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reshape_and_cache", &reshape_and_cache);
   m.def("silu_and_mul", &silu_and_mul);
   m.def("paged_attention_decode", &paged_attention_decode);
+  m.def("paged_attention_decode_split", &paged_attention_decode_split);
   m.def("paged_attention_prefill", &paged_attention_prefill);
   m.def("sgmv_shrink", &sgmv_shrink);
   m.def("sgmv_apply", &sgmv_apply);
