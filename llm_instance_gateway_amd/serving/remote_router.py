@@ -142,6 +142,42 @@ def build_remote_router_app(
             status_code=resp.status_code, content=payload, headers=fwd_headers
         )
 
+    @app.get("/admin/models")
+    async def admin_models():
+        return [
+            {
+                "model_name": m.model_name,
+                "criticality": m.criticality.value,
+                "target_models": [
+                    {"name": t.name, "weight": t.weight} for t in m.targets
+                ],
+                "pool_ref": m.pool_ref,
+            }
+            for m in ds.all_models()
+        ]
+
+    @app.post("/admin/models")
+    async def register_model(body: dict):
+        from ..api.types import Criticality, TargetModel
+
+        ds.store_model(
+            InferenceModel(
+                model_name=str(body["model_name"]),
+                criticality=Criticality(body.get("criticality", "Default")),
+                target_models=[
+                    TargetModel(name=t["name"], weight=int(t.get("weight", 1)))
+                    for t in body.get("target_models", [])
+                ],
+                pool_ref=str(body.get("pool_ref", "")),
+            )
+        )
+        return {"status": "ok"}
+
+    @app.delete("/admin/models/{model_name}")
+    async def delete_model(model_name: str):
+        ds.delete_model(model_name)
+        return {"status": "ok"}
+
     @app.get("/health")
     async def health():
         ok = bool(provider.all_engine_metrics())

@@ -130,3 +130,16 @@ def test_metrics_page_aggregates_fleet(stack):
     assert 'vllm:num_requests_waiting{engine="engine-1"} 90' in text
     assert "gateway_requests_total" in text
     assert client.get("/health").status_code == 200
+
+
+def test_remote_router_admin_models(stack):
+    client, _, _ = stack
+    client.post("/admin/models", json={"model_name": "late-model",
+                                       "criticality": "Critical"})
+    names = [m["model_name"] for m in client.get("/admin/models").json()]
+    assert "late-model" in names
+    r = client.post("/v1/completions", json={"model": "late-model", "prompt": "x"})
+    assert r.status_code == 200
+    client.delete("/admin/models/late-model")
+    names = [m["model_name"] for m in client.get("/admin/models").json()]
+    assert "late-model" not in names
