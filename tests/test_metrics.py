@@ -104,3 +104,43 @@ def test_gateway_metrics_render_all():
         "gateway_e2e_latency_seconds",
     ):
         assert name in text
+
+
+def test_render_parse_roundtrip_property():
+    """Property: render -> parse is lossless for any gauge values and
+    adapter names (incl. escaping-sensitive characters)."""
+    import hypothesis.strategies as st
+    from hypothesis import given, settings
+
+    name_st = st.text(
+        alphabet=st.characters(
+            whitelist_categories=("L", "N"), whitelist_characters="-_./\\\""
+        ),
+        min_size=1,
+        max_size=20,
+    ).filter(lambda s: "," not in s)
+
+    @settings(max_examples=80, deadline=None)
+    @given(
+        adapters=st.lists(name_st, max_size=4, unique=True),
+        running=st.integers(0, 10**6),
+        waiting=st.integers(0, 10**6),
+        kv=st.floats(0, 1, allow_nan=False),
+        max_active=st.integers(0, 64),
+    )
+    def check(adapters, running, waiting, kv, max_active):
+        m = EngineMetrics(
+            active_models={a: 1 for a in adapters},
+            max_active_models=max_active,
+            running_queue_size=running,
+            waiting_queue_size=waiting,
+            kv_cache_usage_percent=kv,
+        )
+        out = parse_engine_metrics(render_engine_metrics("e", m, now=1.0), EngineMetrics())
+        assert out.running_queue_size == running
+        assert out.waiting_queue_size == waiting
+        assert abs(out.kv_cache_usage_percent - kv) < 1e-5
+        assert sorted(out.active_models) == sorted(adapters)
+        assert out.max_active_models == max_active
+
+    check()
