@@ -70,6 +70,49 @@ class GatewayClient:
             body["lora"] = lora
         return self._post("/v1/completions", json=body).json()
 
+    def completion_stream(
+        self,
+        model: str,
+        prompt: Union[str, Sequence[int]],
+        max_tokens: int = 16,
+        temperature: float = 0.0,
+    ):
+        """Yield streamed completion chunks (parsed SSE `data:` events);
+        the final yielded dict carries the usage object."""
+        body = {
+            "model": model,
+            "prompt": list(prompt) if not isinstance(prompt, str) else prompt,
+            "max_tokens": max_tokens,
+            "temperature": temperature,
+            "stream": True,
+        }
+        import json as _json
+
+        url = self.base_url + "/v1/completions"
+        if hasattr(self.session, "stream"):  # httpx / starlette TestClient
+            ctx = self.session.stream("POST", url, json=body)
+            with ctx as resp:
+                self._check(resp)
+                for line in resp.iter_lines():
+                    line = line.decode() if isinstance(line, bytes) else line
+                    if not line.startswith("data: "):
+                        continue
+                    payload = line[len("data: "):]
+                    if payload == "[DONE]":
+                        return
+                    yield _json.loads(payload)
+            return
+        # requests-style session
+        resp = self._check(self.session.post(url, json=body, stream=True))
+        for line in resp.iter_lines():
+            line = line.decode() if isinstance(line, bytes) else line
+            if not line or not line.startswith("data: "):
+                continue
+            payload = line[len("data: "):]
+            if payload == "[DONE]":
+                return
+            yield _json.loads(payload)
+
     def chat(
         self, model: str, messages: List[Dict[str, str]], max_tokens: int = 16
     ) -> Dict[str, Any]:

@@ -55,3 +55,10 @@ def test_client_error_surface(gw):
     with pytest.raises(GatewayError) as ei:
         gw.completion("no-such-model", "x")
     assert ei.value.status == 404
+
+
+def test_client_streaming(gw):
+    chunks = list(gw.completion_stream("m", "go", max_tokens=3))
+    token_chunks = [c for c in chunks if c["choices"][0].get("token") is not None]
+    assert len(token_chunks) == 3
+    assert chunks[-1]["usage"]["completion_tokens"] == 3
