@@ -142,6 +142,17 @@ def build_remote_router_app(
             status_code=resp.status_code, content=payload, headers=fwd_headers
         )
 
+    @app.get("/v1/models")
+    async def list_models():
+        return {
+            "object": "list",
+            "data": [
+                {"id": m.model_name, "object": "model",
+                 "criticality": m.criticality.value}
+                for m in ds.all_models()
+            ],
+        }
+
     @app.get("/admin/models")
     async def admin_models():
         return [

@@ -143,3 +143,9 @@ def test_remote_router_admin_models(stack):
     client.delete("/admin/models/late-model")
     names = [m["model_name"] for m in client.get("/admin/models").json()]
     assert "late-model" not in names
+
+
+def test_remote_router_v1_models(stack):
+    client, _, _ = stack
+    ids = [m["id"] for m in client.get("/v1/models").json()["data"]]
+    assert "my-model" in ids
