@@ -283,8 +283,14 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
             return JSONResponse(status_code=504, content={"error": "request timed out"})
         except RuntimeError as e:
             gw.metrics.requests_total.inc(model=req.model, outcome="error")
-            status = 400 if "max_model_len" in str(e) else 500
-            return JSONResponse(status_code=status, content={"error": str(e)})
+            msg = str(e)
+            if "max_model_len" in msg:
+                status = 400
+            elif "unknown LoRA adapter" in msg:
+                status = 404
+            else:
+                status = 500
+            return JSONResponse(status_code=status, content={"error": msg})
         echo_text = (
             req.prompt if req.echo and isinstance(req.prompt, str) else ""
         )

@@ -281,3 +281,13 @@ def test_overlong_prompt_rejected_400(client):
     )
     assert r.status_code == 400
     assert "max_model_len" in r.json()["error"]
+
+
+def test_unknown_adapter_404(client):
+    r = client.post(
+        "/v1/completions",
+        json={"model": "my-model", "prompt": "x", "max_tokens": 2,
+              "lora": "never-registered"},
+    )
+    assert r.status_code == 404
+    assert "unknown LoRA adapter" in r.json()["error"]
