@@ -107,6 +107,68 @@ def _tp_worker(rank: int, world: int, q):
         ref = full_eng.generate_all()[0].output_token_ids
         out = tp_eng.generate_all()[0].output_token_ids
         assert out == ref, f"rank{rank}: TP tokens {out} != reference {ref}"
+
+        # ---- TP + LoRA numerics: sharded adapter == full adapter ----------
+        # Sharding rules (column-parallel qkv/gate_up: A replicated, B rows
+        # sharded like the base weight; row-parallel o/down: A columns
+        # sharded with the input, B replicated — correct because the SGMV
+        # delta lands BEFORE the block all-reduce).
+        from llm_instance_gateway_amd.lora.manager import AdapterManager
+
+        LR = 8
+        hd, nh, nkv = mc.head_dim, mc.num_heads, mc.num_kv_heads
+        il = mc.intermediate_size
+        g2 = torch.Generator().manual_seed(99)
+
+        def mk(r_, c_):
+            return (torch.randn(r_, c_, generator=g2) * 0.05)
+
+        full_w, tp_w = {}, {}
+        qs, kvs, ils = nh * hd // world, nkv * hd // world, il // world
+        for layer in range(mc.num_layers):
+            for mod in ("qkv", "o", "gate_up", "down"):
+                if mod == "qkv":
+                    A = mk(LR, mc.hidden_size)
+                    B = mk(nh * hd + 2 * nkv * hd, LR)
+                    bq, bk, bv = B.split([nh * hd, nkv * hd, nkv * hd], dim=0)
+                    Bs = torch.cat([
+                        bq[rank * qs:(rank + 1) * qs],
+                        bk[rank * kvs:(rank + 1) * kvs],
+                        bv[rank * kvs:(rank + 1) * kvs],
+                    ], dim=0)
+                    As = A
+                elif mod == "o":
+                    A = mk(LR, nh * hd)
+                    B = mk(mc.hidden_size, LR)
+                    As = A[:, rank * qs:(rank + 1) * qs]
+                    Bs = B
+                elif mod == "gate_up":
+                    A = mk(LR, mc.hidden_size)
+                    B = mk(2 * il, LR)
+                    bg, bu = B.chunk(2, dim=0)
+                    Bs = torch.cat([
+                        bg[rank * ils:(rank + 1) * ils],
+                        bu[rank * ils:(rank + 1) * ils],
+                    ], dim=0)
+                    As = A
+                else:  # down
+                    A = mk(LR, il)
+                    B = mk(mc.hidden_size, LR)
+                    As = A[:, rank * ils:(rank + 1) * ils]
+                    Bs = B
+                full_w[(layer, mod)] = (A, B)
+                tp_w[(layer, mod)] = (As, Bs)
+
+        mgr_f = AdapterManager(full_eng, max_slots=2, rank=LR)
+        mgr_t = AdapterManager(tp_eng, max_slots=2, rank=LR)
+        mgr_f.register_weights("ad", full_w, rank=LR)
+        mgr_t.register_weights("ad", tp_w, rank=LR)
+        full_eng.add_request(prompt, SamplingParams(max_new_tokens=8), lora_name="ad")
+        tp_eng.add_request(prompt, SamplingParams(max_new_tokens=8), lora_name="ad")
+        ref_l = full_eng.generate_all()[0].output_token_ids
+        out_l = tp_eng.generate_all()[0].output_token_ids
+        assert out_l == ref_l, f"rank{rank}: TP+LoRA {out_l} != {ref_l}"
+        assert out_l != ref, "adapter had no effect — test not meaningful"
         q.put((rank, "ok"))
     except Exception as e:  # pragma: no cover
         import traceback
