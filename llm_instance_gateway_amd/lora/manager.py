@@ -144,6 +144,57 @@ class AdapterManager:
                 name=name, rank=r, scaling=self.alpha / r, host_weights=weights
             )
 
+    def register_weights_full(self, name: str, weights: Dict, rank: Optional[int] = None) -> None:
+        """Register an adapter from FULL (unsharded) weights on a TP engine:
+        shards each module's (A, B) to this rank's slice using the base
+        model's layout (column-parallel qkv/gate_up: A replicated, B rows
+        sharded; row-parallel o/down: A columns sharded with the input, B
+        replicated — the SGMV delta lands before the block all-reduce, so
+        the partial sums add up exactly; tests/test_tp.py validates
+        token-identity vs TP=1).  On TP=1 engines this is register_weights."""
+        runner = self.engine.runner
+        tp = runner.tp
+        if tp is None or tp.world_size == 1:
+            self.register_weights(name, weights, rank=rank)
+            return
+        mc = runner.config.model
+        w = tp.world_size
+        r_ = tp.rank
+        hd, nh, nkv = mc.head_dim, mc.num_heads, mc.num_kv_heads
+        qs, kvs = nh * hd // w, nkv * hd // w
+        il = getattr(mc, "intermediate_size", 0)
+        ils = il // w if il else 0
+        sharded: Dict = {}
+        for (layer, mod), (A, B) in weights.items():
+            if mod == "qkv":
+                bq, bk, bv = B.split([nh * hd, nkv * hd, nkv * hd], dim=0)
+                Bs = torch.cat(
+                    [
+                        bq[r_ * qs:(r_ + 1) * qs],
+                        bk[r_ * kvs:(r_ + 1) * kvs],
+                        bv[r_ * kvs:(r_ + 1) * kvs],
+                    ],
+                    dim=0,
+                )
+                As = A
+            elif mod == "o":
+                As = A[:, r_ * qs:(r_ + 1) * qs]
+                Bs = B
+            elif mod == "gate_up":
+                bg, bu = B.chunk(2, dim=0)
+                Bs = torch.cat(
+                    [bg[r_ * ils:(r_ + 1) * ils], bu[r_ * ils:(r_ + 1) * ils]],
+                    dim=0,
+                )
+                As = A
+            elif mod == "down":
+                As = A[:, r_ * ils:(r_ + 1) * ils]
+                Bs = B
+            else:
+                raise KeyError(f"unknown LoRA module {mod!r}")
+            sharded[(layer, mod)] = (As.contiguous(), Bs.contiguous())
+        self.register_weights(name, sharded, rank=rank)
+
     def unregister(self, name: str) -> bool:
         """Sidecar 'unload': evict from GPU and drop from the registry."""
         with self._lock:

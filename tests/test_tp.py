@@ -162,7 +162,13 @@ def _tp_worker(rank: int, world: int, q):
         mgr_f = AdapterManager(full_eng, max_slots=2, rank=LR)
         mgr_t = AdapterManager(tp_eng, max_slots=2, rank=LR)
         mgr_f.register_weights("ad", full_w, rank=LR)
-        mgr_t.register_weights("ad", tp_w, rank=LR)
+        # library-side sharding of FULL weights must equal the hand-sharded
+        # reference slices
+        mgr_t.register_weights_full("ad", full_w, rank=LR)
+        from llm_instance_gateway_amd.lora.manager import _Adapter  # noqa
+        got = mgr_t._adapters["ad"].host_weights
+        for k, (As, Bs) in tp_w.items():
+            assert torch.equal(got[k][0], As) and torch.equal(got[k][1], Bs), k
         full_eng.add_request(prompt, SamplingParams(max_new_tokens=8), lora_name="ad")
         tp_eng.add_request(prompt, SamplingParams(max_new_tokens=8), lora_name="ad")
         ref_l = full_eng.generate_all()[0].output_token_ids
