@@ -245,10 +245,11 @@ def _worker_main(
                 if mgr is None:
                     raise RuntimeError("engine has no adapter slots")
                 if isinstance(source, dict):
-                    mgr.register_weights(name, source, rank=rank)
+                    mgr.register_weights_full(name, source, rank=rank)
                 elif isinstance(source, str) and os.path.exists(source):
                     weights = torch.load(source, weights_only=True)
-                    mgr.register_weights(name, weights, rank=rank)
+                    # full (unsharded) checkpoints: TP engines shard in-place
+                    mgr.register_weights_full(name, weights, rank=rank)
                 else:
                     mgr.register_random(
                         name, seed=abs(hash(name)) % (2**31), rank=rank
