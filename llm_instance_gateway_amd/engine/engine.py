@@ -152,6 +152,7 @@ class LLMEngine:
             waiting_queue_size=self.scheduler.num_waiting,
             kv_cache_usage_percent=self.scheduler.kv_usage,
             kv_cache_max_token_capacity=self.config.kv_tokens_capacity,
+            num_preemptions_total=self.scheduler.total_preemptions,
         )
 
     # -- convenience: run to completion (tests / offline batch) -----------------

@@ -62,6 +62,8 @@ class EngineScheduler:
         )
         self.waiting: Deque[Sequence] = deque()
         self.running: List[Sequence] = []
+        # lifetime preemption count (vllm:num_preemptions_total parity)
+        self.total_preemptions = 0
 
     # -- intake --------------------------------------------------------------
     def add_seq(self, seq: Sequence) -> None:
@@ -137,6 +139,7 @@ class EngineScheduler:
         self.running.remove(victim)
         self.block_manager.free(victim)
         victim.preempt()
+        self.total_preemptions += 1
         self.waiting.appendleft(victim)
         out.preempted.append(victim)
         return True
@@ -175,6 +178,7 @@ class EngineScheduler:
                     self.running.remove(seq)
                     bm.free(seq)
                     seq.preempt()
+                    self.total_preemptions += 1
                     self.waiting.appendleft(seq)
                     out.preempted.append(seq)
                     break

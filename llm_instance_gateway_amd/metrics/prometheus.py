@@ -39,6 +39,7 @@ LORA_INFO_MAX_LABEL = "max_lora"
 RUNNING_QUEUE_METRIC = "vllm:num_requests_running"
 WAITING_QUEUE_METRIC = "vllm:num_requests_waiting"
 KV_CACHE_USAGE_METRIC = "vllm:gpu_cache_usage_perc"
+PREEMPTIONS_METRIC = "vllm:num_preemptions_total"
 
 
 def _escape_label(v: str) -> str:
@@ -63,6 +64,8 @@ def render_engine_metrics(
         f"{WAITING_QUEUE_METRIC}{lab} {m.waiting_queue_size}",
         f"# TYPE {KV_CACHE_USAGE_METRIC} gauge",
         f"{KV_CACHE_USAGE_METRIC}{lab} {m.kv_cache_usage_percent:.6f}",
+        f"# TYPE {PREEMPTIONS_METRIC} counter",
+        f"{PREEMPTIONS_METRIC}{lab} {m.num_preemptions_total}",
         f"# TYPE {LORA_INFO_METRIC} gauge",
         "{}{{engine=\"{}\",{}=\"{}\",{}=\"{}\"}} {:.3f}".format(
             LORA_INFO_METRIC,
@@ -118,6 +121,7 @@ def parse_engine_metrics(text: str, existing: EngineMetrics) -> EngineMetrics:
         waiting_queue_size=existing.waiting_queue_size,
         kv_cache_usage_percent=existing.kv_cache_usage_percent,
         kv_cache_max_token_capacity=existing.kv_cache_max_token_capacity,
+        num_preemptions_total=existing.num_preemptions_total,
     )
     latest_lora: Optional[Tuple[float, Dict[str, str]]] = None
     for raw in text.splitlines():
@@ -131,6 +135,8 @@ def parse_engine_metrics(text: str, existing: EngineMetrics) -> EngineMetrics:
             out.waiting_queue_size = int(value)
         elif name == KV_CACHE_USAGE_METRIC:
             out.kv_cache_usage_percent = value
+        elif name == PREEMPTIONS_METRIC:
+            out.num_preemptions_total = int(value)
         elif name == LORA_INFO_METRIC:
             # latest-wins by gauge value as timestamp (metrics.go:135-150)
             if latest_lora is None or value >= latest_lora[0]:

@@ -53,6 +53,9 @@ class EngineMetrics:
     waiting_queue_size: int = 0
     kv_cache_usage_percent: float = 0.0
     kv_cache_max_token_capacity: int = 0
+    # lifetime preemption count (vLLM exports vllm:num_preemptions_total;
+    # the reference sim tracks the same as recompute_count)
+    num_preemptions_total: int = 0
 
 
 @dataclass
@@ -77,6 +80,7 @@ class EngineState:
                 waiting_queue_size=m.waiting_queue_size,
                 kv_cache_usage_percent=m.kv_cache_usage_percent,
                 kv_cache_max_token_capacity=m.kv_cache_max_token_capacity,
+                num_preemptions_total=m.num_preemptions_total,
             ),
         )
 

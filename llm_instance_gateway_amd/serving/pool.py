@@ -51,7 +51,8 @@ _SLOT_KV_USAGE = 4
 _SLOT_KV_CAP = 5
 _SLOT_MAX_LORAS = 6
 _SLOT_DRAINING = 7
-SLOTS_PER_ENGINE = 8
+_SLOT_PREEMPTIONS = 8
+SLOTS_PER_ENGINE = 9
 
 HEARTBEAT_STALE_S = 30.0
 
@@ -186,6 +187,7 @@ def _worker_main(
         shared[b + _SLOT_KV_CAP] = float(m.kv_cache_max_token_capacity)
         shared[b + _SLOT_MAX_LORAS] = float(m.max_active_models)
         shared[b + _SLOT_DRAINING] = 1.0 if draining else 0.0
+        shared[b + _SLOT_PREEMPTIONS] = float(m.num_preemptions_total)
 
     def send(msg) -> None:
         if is_front:
@@ -519,6 +521,7 @@ class EngineHandle:
             waiting_queue_size=int(s[b + _SLOT_WAITING]),
             kv_cache_usage_percent=s[b + _SLOT_KV_USAGE],
             kv_cache_max_token_capacity=int(s[b + _SLOT_KV_CAP]),
+            num_preemptions_total=int(s[b + _SLOT_PREEMPTIONS]),
         )
 
     def heartbeat_age_s(self) -> float:

@@ -144,3 +144,9 @@ def test_render_parse_roundtrip_property():
         assert out.max_active_models == max_active
 
     check()
+
+
+def test_preemption_counter_roundtrip():
+    m = EngineMetrics(num_preemptions_total=17)
+    out = parse_engine_metrics(render_engine_metrics("e", m, now=1.0), EngineMetrics())
+    assert out.num_preemptions_total == 17
