@@ -84,3 +84,42 @@ def test_snapshot_isolated_from_updates():
     c = snap.clone()
     c.metrics.waiting_queue_size = 123
     assert snap.metrics.waiting_queue_size == 0
+
+
+def test_refresh_loops_update_live():
+    """The reference's two daemon refresh loops (provider.go:60-101): with a
+    mutating source, the snapshot follows within a few intervals."""
+    import time as _t
+
+    from llm_instance_gateway_amd.testing import FakeMetricsSource
+
+    ds = Datastore(engines=[Engine("e0", 0)])
+    src = FakeMetricsSource(res={"e0": EngineMetrics(waiting_queue_size=1)})
+    p = Provider(src, ds)
+    p.init(refresh_engines_interval_s=0.05, refresh_metrics_interval_s=0.02)
+    try:
+        deadline = _t.monotonic() + 5
+        while _t.monotonic() < deadline:
+            states = p.all_engine_metrics()
+            if states and states[0].metrics.waiting_queue_size == 1:
+                break
+            _t.sleep(0.02)
+        assert states[0].metrics.waiting_queue_size == 1
+        # source moves; snapshot follows
+        src.res["e0"] = EngineMetrics(waiting_queue_size=9)
+        deadline = _t.monotonic() + 5
+        while _t.monotonic() < deadline:
+            if p.all_engine_metrics()[0].metrics.waiting_queue_size == 9:
+                break
+            _t.sleep(0.02)
+        assert p.all_engine_metrics()[0].metrics.waiting_queue_size == 9
+        # engine-set loop follows the datastore
+        ds.add_engine(Engine("e1", 1))
+        deadline = _t.monotonic() + 5
+        while _t.monotonic() < deadline:
+            if len(p.all_engine_metrics()) == 2:
+                break
+            _t.sleep(0.02)
+        assert len(p.all_engine_metrics()) == 2
+    finally:
+        p.stop()
