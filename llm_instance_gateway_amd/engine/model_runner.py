@@ -184,6 +184,7 @@ class ModelRunner:
                 if decodes
                 else None
             ),
+            max_decode_ctx=max(d_ctx) if d_ctx else 0,
         )
         # Per-token prefill expansion, once per step (the attention layers of
         # every transformer block reuse it; see profiles/r01_bench_kernel_stats.md).
