@@ -67,6 +67,11 @@ def build_remote_router_app(
     app.state.datastore = ds
     app.state.provider = provider
 
+    @app.on_event("shutdown")
+    async def _shutdown():
+        provider.stop()
+        await client.aclose()
+
     @app.post("/v1/completions")
     async def completions(req: CompletionRequest):
         t0 = time.monotonic()
