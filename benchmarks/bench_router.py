@@ -67,6 +67,9 @@ def main() -> int:
                     help="registered InferenceModels round-robined by requests")
     ap.add_argument("--requests", type=int, default=100_000)
     ap.add_argument("--sheddable-frac", type=float, default=0.2)
+    ap.add_argument("--native", action="store_true",
+                    help="use the C++ filter tree (state synced once: the "
+                    "fleet is static in this harness)")
     args = ap.parse_args()
 
     provider = FakeProvider(args.engines, args.adapters_per_engine)
@@ -80,7 +83,14 @@ def main() -> int:
                 target_models=[TargetModel(name=target, weight=100)],
             )
         )
-    sched = Scheduler(provider)
+    if args.native:
+        from llm_instance_gateway_amd.router.native import NativeScheduler, native_available
+
+        assert native_available(), "_router extension not built"
+        sched = NativeScheduler(seed=1)
+        sched.sync_states(provider.all_engine_metrics())
+    else:
+        sched = Scheduler(provider)
 
     lat = []
     shed = 0
@@ -104,6 +114,7 @@ def main() -> int:
 
     lat_us = sorted(x * 1e6 for x in lat)
     out = {
+        "router": "native" if args.native else "python",
         "requests": args.requests,
         "engines": args.engines,
         "models": args.models,

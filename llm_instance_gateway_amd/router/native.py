@@ -1,0 +1,161 @@
+"""Python wrapper for the native (C++) filter-tree router.
+
+`NativeScheduler` is a drop-in for `router.scheduler.Scheduler` backed by
+`_router.NativeRouter` (router/csrc/router_native.cpp).  Two sync modes:
+
+  * **pool mode** (`NativeScheduler.for_pool`): gauges are read by the C++
+    tree DIRECTLY from the engine pool's shared-memory array (zero staleness,
+    zero per-request copying); only the slow-changing state (engine set,
+    resident-adapter sets) syncs lazily on change.
+  * **state mode** (`sync_states`): push EngineState snapshots (remote-scrape
+    deployments, tests).
+
+Falls back cleanly: `native_available()` is False when the extension is not
+built (pure-Python installs keep working on `Scheduler`).
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+from .scheduler import ResourceExhausted, SchedulingError
+from .types import Engine, EngineState, LLMRequest
+
+try:
+    from llm_instance_gateway_amd import _router as _ext
+except ImportError:  # pragma: no cover
+    _ext = None
+
+
+def native_available() -> bool:
+    return _ext is not None
+
+
+class NativeScheduler:
+    def __init__(
+        self,
+        seed: int = 0,
+        kv_cache_threshold: float = 0.8,
+        queue_threshold_critical: int = 5,
+        queueing_threshold_lora: int = 50,
+    ):
+        if _ext is None:  # pragma: no cover
+            raise RuntimeError("native router extension _router is not built")
+        self._r = _ext.NativeRouter(seed)
+        self._r.set_thresholds(
+            kv_cache_threshold, queue_threshold_critical, queueing_threshold_lora
+        )
+        self._pool = None
+        self._provider = None
+        self._provider_version = -1
+        self._sync_cache = {}
+        self._state_names: set = set()
+        self._engines_by_name = {}
+
+    # -- pool mode -----------------------------------------------------------
+    @classmethod
+    def for_pool(cls, pool, seed: int = 0, **thresholds) -> "NativeScheduler":
+        from ..serving import pool as pool_mod
+
+        self = cls(seed=seed, **thresholds)
+        self._pool = pool
+        self._r.attach_shm(
+            memoryview(pool.shared),
+            pool_mod.SLOTS_PER_ENGINE,
+            pool_mod._SLOT_WAITING,
+            pool_mod._SLOT_KV_USAGE,
+            pool_mod._SLOT_DRAINING,
+        )
+        return self
+
+    @classmethod
+    def for_provider(cls, provider, seed: int = 0, **thresholds) -> "NativeScheduler":
+        """Mirror a refresh-mode Provider: re-sync only when its version
+        bumps (per scrape round), schedule natively in between."""
+        self = cls(seed=seed, **thresholds)
+        self._provider = provider
+        return self
+
+    def _sync_pool(self) -> None:
+        pool = self._pool
+        seen = set()
+        for name, h in pool.handles.items():
+            seen.add(name)
+            if not h.healthy or pool.datastore.get_engine(name) is None:
+                if self._sync_cache.pop(name, None) is not None:
+                    self._r.remove(name)
+                continue
+            key = (tuple(h.active_adapters), h.max_slots, h.engine.pool)
+            if self._sync_cache.get(name) != key:
+                self._sync_cache[name] = key
+                self._r.upsert(
+                    name,
+                    pool=h.engine.pool,
+                    slot_base=h.slot_base,
+                    adapters=list(h.active_adapters),
+                    max_active=h.max_slots,
+                )
+        for name in list(self._sync_cache):
+            if name not in seen:
+                del self._sync_cache[name]
+                self._r.remove(name)
+
+    # -- state mode ----------------------------------------------------------
+    def sync_states(self, states: List[EngineState]) -> None:
+        seen = set()
+        for s in states:
+            seen.add(s.engine.name)
+            self._engines_by_name[s.engine.name] = s.engine
+            m = s.metrics
+            self._r.upsert(
+                s.engine.name,
+                pool=s.engine.pool,
+                slot_base=-1,
+                adapters=list(m.active_models),
+                max_active=m.max_active_models,
+            )
+            self._r.set_gauges(
+                s.engine.name,
+                float(m.waiting_queue_size),
+                float(m.running_queue_size),
+                float(m.kv_cache_usage_percent),
+                False,
+            )
+        # drop engines that vanished since the last sync
+        for name in self._state_names - seen:
+            self._r.remove(name)
+            self._engines_by_name.pop(name, None)
+        self._state_names = seen
+
+    # -- the Scheduler interface ----------------------------------------------
+    def schedule(self, req: LLMRequest) -> Engine:
+        if self._pool is not None:
+            self._sync_pool()
+        elif self._provider is not None:
+            v = self._provider.version
+            if v != self._provider_version:
+                self._provider_version = v
+                self.sync_states(self._provider.all_engine_metrics())
+        try:
+            name = self._r.schedule(
+                req.resolved_target_model, bool(req.critical), req.pool_ref or ""
+            )
+        except RuntimeError as e:
+            raise SchedulingError(str(e)) from e
+        if name == "":
+            raise ResourceExhausted(
+                "dropping request due to limited backend resources"
+            )
+        if self._pool is not None:
+            eng = self._pool.datastore.get_engine(name)
+            if eng is not None:
+                return eng
+        return self._engines_by_name.get(name) or Engine(name=name)
+
+    def survivors(self, req: LLMRequest) -> List[str]:
+        """Survivor names after the tree (equivalence tests)."""
+        return list(
+            self._r.survivors(
+                req.resolved_target_model, bool(req.critical), req.pool_ref or ""
+            )
+        )

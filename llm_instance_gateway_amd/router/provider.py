@@ -51,6 +51,9 @@ class Provider:
         self._datastore = datastore
         self._lock = threading.Lock()
         self._states: Dict[str, EngineState] = {}
+        # bumped on every state change: cheap staleness check for consumers
+        # that mirror the snapshot (router/native.py syncs only on change)
+        self.version = 0
         self._stop = threading.Event()
         self._threads: List[threading.Thread] = []
 
@@ -77,6 +80,7 @@ class Provider:
         mode the engine worker calls this after every scheduler iteration."""
         with self._lock:
             self._states[engine.name] = EngineState(engine=engine, metrics=metrics)
+            self.version += 1
 
     # -- lifecycle (provider.go:60-101) -------------------------------------
     def init(
@@ -124,12 +128,17 @@ class Provider:
         add new engines with empty metrics, drop vanished ones."""
         engines = {e.name: e for e in self._datastore.get_engines()}
         with self._lock:
+            changed = False
             for name, eng in engines.items():
                 if name not in self._states:
                     self._states[name] = EngineState(engine=eng)
+                    changed = True
             for name in list(self._states):
                 if name not in engines:
                     del self._states[name]
+                    changed = True
+            if changed:
+                self.version += 1
 
     def refresh_metrics_once(self, timeout_s: float = DEFAULT_FETCH_TIMEOUT_S) -> List[Exception]:
         """Fan-out fetch for every engine (provider.go:134-179).  Errors are
@@ -163,6 +172,8 @@ class Provider:
                     self._states[name] = EngineState(
                         engine=self._states[name].engine, metrics=metrics
                     )
+            if results:
+                self.version += 1
         if errs:
             log.debug("refresh_metrics_once: %d fetch errors", len(errs))
         return errs

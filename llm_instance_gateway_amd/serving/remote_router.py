@@ -59,7 +59,13 @@ def build_remote_router_app(
         refresh_metrics_interval_s=refresh_metrics_interval_s,
         start_loops=start_refresh,
     )
-    scheduler = Scheduler(provider)
+    from ..router.native import NativeScheduler, native_available
+
+    if native_available():
+        scheduler = NativeScheduler.for_provider(provider)
+        log.info("remote-router using the native (C++) filter tree")
+    else:  # pragma: no cover - extension always built in-tree
+        scheduler = Scheduler(provider)
     gw_metrics = GatewayMetrics()
     client = httpx.AsyncClient(timeout=request_timeout_s)
 

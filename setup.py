@@ -13,10 +13,13 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
+import pybind11
+from setuptools import Extension
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
 ROOT = Path(__file__).parent
 CSRC = ROOT / "llm_instance_gateway_amd" / "ops" / "csrc"
+RCSRC = ROOT / "llm_instance_gateway_amd" / "router" / "csrc"
 
 sources = sorted(
     str(p)
@@ -38,10 +41,19 @@ ext = CUDAExtension(
     },
 )
 
+# pure-C++ native router (no GPU/torch dependency): plain pybind11 module
+router_ext = Extension(
+    name="llm_instance_gateway_amd._router",
+    sources=[str(RCSRC / "router_native.cpp")],
+    include_dirs=[pybind11.get_include()],
+    extra_compile_args=["-O3", "-std=c++17"],
+    language="c++",
+)
+
 setup(
     name="llm_instance_gateway_amd",
     version="0.1.0",
     packages=["llm_instance_gateway_amd"],
-    ext_modules=[ext],
+    ext_modules=[ext, router_ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )
