@@ -136,3 +136,49 @@ def test_native_reads_pool_shared_memory():
     # draining flag in shm removes an engine from consideration
     p.shared[pool_mod.SLOTS_PER_ENGINE * 1 + pool_mod._SLOT_DRAINING] = 1.0
     assert ns.schedule(LLMRequest(model="m", critical=True)).name == "engine-0"
+
+
+def test_native_for_pool_end_to_end():
+    """Real engine pool + native router: route, dispatch, adapter-affinity
+    sync on adapter load, drain visibility through shared memory."""
+    from llm_instance_gateway_amd.api.types import InferenceModel
+    from llm_instance_gateway_amd.serving import EnginePool, PoolRequest
+
+    p = EnginePool(
+        "tiny",
+        devices=["cpu", "cpu"],
+        engine_kwargs=dict(
+            num_kv_blocks=64, max_num_seqs=4,
+            max_num_batched_tokens=128, max_model_len=128, max_loras=2,
+        ),
+        lora_rank=8,
+    )
+    p.start(watchdog=False)
+    try:
+        p.datastore.store_model(InferenceModel(model_name="m"))
+        ns = NativeScheduler.for_pool(p, seed=4)
+        p.scheduler = ns  # route through the native tree
+        eng = p.route(LLMRequest(model="m", critical=True))
+        assert eng.name in p.handles
+        t = p.route_and_submit(
+            LLMRequest(model="m", critical=True),
+            PoolRequest(request_id="nr-1", prompt_token_ids=[1, 2, 3],
+                        max_new_tokens=2),
+        )
+        assert len(t.wait(60).output_token_ids) == 2
+        # adapter residency flows into native affinity
+        assert p.handles["engine-0"].load_adapter("nat-ad") is None
+        import time as _t
+
+        deadline = _t.monotonic() + 5
+        while _t.monotonic() < deadline:
+            if "nat-ad" in p.handles["engine-0"].active_adapters:
+                break
+            _t.sleep(0.02)
+        for _ in range(6):
+            eng = ns.schedule(
+                LLMRequest(model="m", resolved_target_model="nat-ad", critical=True)
+            )
+            assert eng.name == "engine-0"  # affinity wins
+    finally:
+        p.stop()
