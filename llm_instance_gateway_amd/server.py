@@ -74,6 +74,11 @@ def main(argv: Optional[List[str]] = None) -> int:
         help="comma list of host:port model servers — remote-router mode "
         "(the reference's topology: scrape + schedule + forward, no local GPUs)",
     )
+    ap.add_argument(
+        "--native-router", action="store_true",
+        help="route with the C++ filter tree reading shared-memory gauges "
+        "(default: Python tree; both are semantically identical)",
+    )
     ap.add_argument("--v", type=int, default=1, help="log verbosity (klog analogue)")
     args = ap.parse_args(argv)
 
@@ -111,6 +116,22 @@ def main(argv: Optional[List[str]] = None) -> int:
     pool = build_pool(args, cfg)
     log.info("starting %d engine(s)...", len(pool.handles))
     pool.start()
+
+    if args.native_router:
+        from .router.native import NativeScheduler, native_available
+
+        if native_available():
+            kw = {}
+            if cfg:
+                kw = dict(
+                    kv_cache_threshold=cfg.kv_cache_threshold,
+                    queue_threshold_critical=cfg.queue_threshold_critical,
+                    queueing_threshold_lora=cfg.queueing_threshold_lora,
+                )
+            pool.scheduler = NativeScheduler.for_pool(pool, **kw)
+            log.info("routing with the native (C++) filter tree")
+        else:  # pragma: no cover
+            log.warning("--native-router requested but _router is not built")
 
     reconciler = None
     if args.config:
