@@ -210,14 +210,29 @@ class EngineScheduler:
             out.scheduled.append(ScheduledSeq(seq, chunk))
             token_budget -= chunk
 
-        # 2) Admit waiting sequences (FIFO) under caps.
+        # 2) Admit waiting sequences (FIFO) under caps.  A batch may not need
+        # more DISTINCT adapters than the engine has slots (vLLM's curr_loras
+        # rule): adapter-blocked sequences are SKIPPED (not head-of-line
+        # blocking — a later request with a resident adapter may proceed).
+        adapters_in_use = {
+            s.lora_name for s in self.running if s.lora_name
+        }
+        skip = 0
         while (
-            self.waiting
+            skip < len(self.waiting)
             and token_budget > 0
             and len(self.running) < cfg.max_num_seqs
             and bm.usage < cfg.preempt_kv_threshold
         ):
-            seq = self.waiting[0]
+            seq = self.waiting[skip]
+            if (
+                seq.lora_name
+                and cfg.max_loras > 0
+                and seq.lora_name not in adapters_in_use
+                and len(adapters_in_use) >= cfg.max_loras
+            ):
+                skip += 1  # adapter slots exhausted for this batch
+                continue
             # Recompute path included: a preempted seq re-prefills prompt +
             # prior output (next_chunk_tokens covers total_len).
             chunk = seq.next_chunk_tokens(token_budget)
@@ -226,7 +241,9 @@ class EngineScheduler:
             target = seq.num_computed_tokens + chunk
             if not bm.can_grow(seq, target, respect_watermark=True):
                 break  # head-of-line blocking, same as vLLM FIFO admission
-            self.waiting.popleft()
+            del self.waiting[skip]
+            if seq.lora_name:
+                adapters_in_use.add(seq.lora_name)
             bm.grow(seq, target)
             seq.status = SequenceStatus.RUNNING
             if seq.first_scheduled_time is None:
