@@ -132,7 +132,9 @@ class Gateway:
         lreq = LLMRequest(
             model=req.model,
             target_models={t.name: t.weight for t in im.targets},
-            resolved_target_model=target,
+            # affinity routing keys on the adapter actually being used: an
+            # explicit `lora` override supersedes the weighted draw
+            resolved_target_model=lora or target,
             critical=critical,
             prompt_len=len(prompt_ids),
             max_new_tokens=req.max_tokens,

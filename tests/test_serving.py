@@ -291,3 +291,27 @@ def test_unknown_adapter_404(client):
     )
     assert r.status_code == 404
     assert "unknown LoRA adapter" in r.json()["error"]
+
+
+def test_concurrent_adapter_churn_through_frontdoor(client, pool):
+    """Config #5's churn at the serving layer: concurrent requests over more
+    adapters than slots (max_loras=4/engine) through the real front door."""
+    import concurrent.futures
+
+    for i in range(6):
+        client.post("/v1/load_lora_adapter", json={"lora_name": f"churn-{i}"})
+
+    def one(i):
+        return client.post(
+            "/v1/completions",
+            json={"model": "my-model", "prompt": f"req {i}", "max_tokens": 2,
+                  "lora": f"churn-{i % 6}"},
+        )
+
+    with concurrent.futures.ThreadPoolExecutor(max_workers=6) as ex:
+        results = list(ex.map(one, range(12)))
+    # adapters are registered pool-wide: every request must complete, with
+    # slot churn (6 adapters > 4 slots) handled by LRU evict + reload
+    for r in results:
+        assert r.status_code == 200, r.text
+        assert len(r.json()["choices"][0]["token_ids"]) == 2
