@@ -171,3 +171,57 @@ def test_adapter_churn_under_load():
         )
     eng2.generate_all(max_steps=500)
     assert [s.output_token_ids for s in seqs] == [s.output_token_ids for s in seqs2]
+
+
+def test_preemption_with_adapter_eviction_interplay():
+    """A preempted sequence whose adapter got LRU-evicted meanwhile must
+    reload it on re-admission (prepare() per step) and still finish with the
+    right weights (vs an uncontended run)."""
+    import torch
+
+    from llm_instance_gateway_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from llm_instance_gateway_amd.lora.manager import AdapterManager
+    from llm_instance_gateway_amd.models.config import get_model_config
+
+    def build(num_kv_blocks, max_loras):
+        torch.manual_seed(0)
+        eng = LLMEngine(
+            EngineConfig(
+                model=get_model_config("tiny"),
+                num_kv_blocks=num_kv_blocks,
+                max_num_seqs=4,
+                max_num_batched_tokens=64,
+                max_model_len=192,
+                max_loras=max_loras,
+                watermark_blocks=0,  # allow overcommit so preemption fires
+            ),
+            device="cpu",
+        )
+        mgr = AdapterManager(eng, max_slots=max_loras, rank=8)
+        for i in range(3):
+            mgr.register_random(f"pe-{i}", seed=50 + i)
+        return eng
+
+    def run(eng):
+        g = torch.Generator().manual_seed(2)
+        seqs = []
+        for i in range(4):
+            prompt = torch.randint(1, 500, (40,), generator=g).tolist()
+            seqs.append(
+                eng.add_request(
+                    prompt, SamplingParams(max_new_tokens=30),
+                    lora_name=f"pe-{i % 3}",
+                )
+            )
+        eng.generate_all(max_steps=2000)
+        return [s.output_token_ids for s in seqs], seqs
+
+    # tiny KV pool + tight slots: preemption AND eviction both fire
+    contended = build(num_kv_blocks=11, max_loras=2)
+    out_c, seqs_c = run(contended)
+    assert sum(s.num_preemptions for s in seqs_c) > 0, "no preemption exercised"
+
+    # roomy pool, all adapters resident: the reference answer
+    roomy = build(num_kv_blocks=128, max_loras=3)
+    out_r, _ = run(roomy)
+    assert out_c == out_r
