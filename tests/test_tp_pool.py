@@ -95,3 +95,42 @@ def test_tp_engine_adapter_load(mixed_pool):
     while time.monotonic() < deadline and "tp-adapter" not in h.active_adapters:
         time.sleep(0.02)
     assert "tp-adapter" in h.active_adapters
+
+
+def test_tp_engine_follower_death_detected_and_restarted():
+    """A dead TP FOLLOWER rank must take the whole engine out of routing
+    (the front rank's broadcast would hang); restart re-spawns the group."""
+    p = EnginePool(
+        specs=[
+            EngineSpec(
+                name="tp-x", model="tiny", devices=["cpu", "cpu"],
+                engine_kwargs=EKW, lora_rank=8,
+            )
+        ]
+    )
+    p.start()  # watchdog on
+    try:
+        h = p.handles["tp-x"]
+        assert h.alive() and len(h.procs) == 2
+        h.procs[1].terminate()  # kill the follower
+        deadline = time.monotonic() + 15
+        while time.monotonic() < deadline:
+            if not p.all_engine_metrics():
+                break
+            time.sleep(0.05)
+        assert not p.all_engine_metrics(), "dead TP engine still routable"
+        p.restart_engine("tp-x")
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            if p.handles["tp-x"].alive():
+                break
+            time.sleep(0.05)
+        assert p.handles["tp-x"].alive()
+        from llm_instance_gateway_amd.serving import PoolRequest
+
+        t = p.submit("tp-x", PoolRequest(request_id="tpr-1",
+                                         prompt_token_ids=[1, 2, 3],
+                                         max_new_tokens=2))
+        assert len(t.wait(60).output_token_ids) == 2
+    finally:
+        p.stop()
