@@ -191,3 +191,77 @@ def test_load_config_file(tmp_path):
     p.write_text(CFG_YAML)
     cfg = load_config(str(p))
     assert cfg.pool.base_model == "tiny"
+
+
+def test_reconcile_flips_under_live_traffic(tmp_path, pool):
+    """Adapter rollout flips (ensureExist <-> ensureNotExist) while requests
+    stream through the engine: no crashes, traffic keeps completing, final
+    registry state matches the last config."""
+    import threading
+
+    from llm_instance_gateway_amd.serving import PoolRequest
+
+    path = tmp_path / "flip.yaml"
+
+    def write(exist, not_exist):
+        path.write_text(
+            "pool: {name: p, base_model: tiny}\n"
+            "models: [{model_name: flip-model}]\n"
+            f"ensureExist: {{models: [{', '.join('{id: %s, rank: 8}' % a for a in exist)}]}}\n"
+            f"ensureNotExist: {{models: [{', '.join('{id: %s}' % a for a in not_exist)}]}}\n"
+        )
+
+    write(["flip-a", "flip-b"], [])
+    rec = ConfigReconciler(
+        str(path), pool, poll_interval_s=0.05,
+        health_timeout_s=5, health_interval_s=0.05,
+    )
+    rec.start()
+    stop = threading.Event()
+    errors = []
+    done_count = [0]
+
+    def traffic():
+        i = 0
+        while not stop.is_set():
+            i += 1
+            h = pool.handles["engine-0"]
+            if "flip-a" not in h.adapter_registry:
+                time.sleep(0.01)
+                continue
+            try:
+                t = pool.submit(
+                    "engine-0",
+                    PoolRequest(request_id=f"flip-{i}",
+                                prompt_token_ids=[1, 2, 3, 4],
+                                max_new_tokens=2, lora_name="flip-a"),
+                )
+                t.wait(30)
+                done_count[0] += 1
+            except Exception as e:  # pragma: no cover
+                errors.append(e)
+                return
+
+    th = threading.Thread(target=traffic, daemon=True)
+    th.start()
+    try:
+        for round_ in range(3):
+            time.sleep(0.15)
+            write(["flip-a"], ["flip-b"])
+            time.sleep(0.15)
+            write(["flip-a", "flip-b"], [])
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline and done_count[0] < 3:
+            time.sleep(0.05)
+        # wait for the reconciler to apply the FINAL config before stopping
+        h = pool.handles["engine-0"]
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline and "flip-b" not in h.adapter_registry:
+            time.sleep(0.05)
+    finally:
+        stop.set()
+        th.join(timeout=30)
+        rec.stop()
+    assert not errors, errors[:1]
+    assert done_count[0] >= 3
+    assert {"flip-a", "flip-b"} <= set(h.adapter_registry)
