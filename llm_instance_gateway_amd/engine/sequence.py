@@ -37,6 +37,9 @@ class SamplingParams:
     seed: int = 0
     # OpenAI `stop`: token-id sequences; generation halts when the output
     # tail matches one, and the matched tail is trimmed from the output.
+    # Token-id matching equals string matching under the front door's
+    # byte-level codec (utils/tokenizer.py: one byte == one id, context-free);
+    # a subword tokenizer would need decoded-text matching instead.
     stop_sequences: List[List[int]] = field(default_factory=list)
 
 

@@ -351,6 +351,16 @@ class AdapterManager:
                     tile_t0, meta.lora_tile_len, meta.lora_tile_slot, 1.0,
                     ops._stream(),
                 )
+                # tiles cover only the prefill partition; a mixed batch's
+                # decode suffix still needs the per-token path or those
+                # tokens silently lose their LoRA delta
+                np_ = int(getattr(meta, "num_prefill_tokens", 0))
+                if np_ < T:
+                    ops._EXT.sgmv_apply(
+                        y[np_:], x[np_:], self.a_stacks[key],
+                        self.b_stacks[key], ws, slots[np_:],
+                        1.0, False, ops._stream(),
+                    )
             else:
                 # decode (or odd dims): per-token kernels; wide variants only
                 # help when consecutive tokens share an adapter

@@ -219,20 +219,21 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
         rid = ticket.request_id
 
         if req.stream:
+            # OpenAI semantics: a matched stop sequence is never streamed.
+            # Hold back a tail of len(longest stop) tokens until the stop
+            # check clears them; at stream end flush from the engine's
+            # trimmed output so matched stop tokens are dropped.
+            stop_lists = [tokenizer.encode(s, add_bos=False)
+                          for s in ([req.stop] if isinstance(req.stop, str)
+                                    else (req.stop or [])) if s]
+            holdback = max((len(s) for s in stop_lists), default=0)
+
             async def sse() -> AsyncIterator[str]:
                 loop = asyncio.get_running_loop()
                 sent = 0
-                while True:
-                    try:
-                        tok = await loop.run_in_executor(None, ticket.tokens.get)
-                    except asyncio.CancelledError:
-                        # client disconnected mid-stream: stop decoding for a
-                        # dead connection (engine abort + block free)
-                        pool.abort(rid)
-                        raise
-                    if tok is None:
-                        break
-                    sent += 1
+                pending: List[int] = []
+
+                def chunk_for(tok: int) -> str:
                     chunk = {
                         "id": rid,
                         "object": "text_completion",
@@ -243,16 +244,36 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
                              "token": tok, "finish_reason": None}
                         ],
                     }
-                    yield f"data: {json.dumps(chunk)}\n\n"
+                    return f"data: {json.dumps(chunk)}\n\n"
+
+                while True:
+                    try:
+                        tok = await loop.run_in_executor(None, ticket.tokens.get)
+                    except asyncio.CancelledError:
+                        # client disconnected mid-stream: stop decoding for a
+                        # dead connection (engine abort + block free)
+                        pool.abort(rid)
+                        raise
+                    if tok is None:
+                        break
+                    pending.append(tok)
+                    while len(pending) > holdback:
+                        sent += 1
+                        yield chunk_for(pending.pop(0))
                 await _await_ticket(ticket)
                 gw.record_finished(ticket, req.model)
                 fin = ticket.result
+                if fin is not None:
+                    # flush survivors of the stop trim (held-back tail)
+                    for tok in fin.output_token_ids[sent:]:
+                        yield chunk_for(tok)
                 final = {
                     "id": rid,
                     "object": "text_completion",
                     "created": created,
                     "model": req.model,
-                    "choices": [{"index": 0, "text": "", "finish_reason": "length"}],
+                    "choices": [{"index": 0, "text": "",
+                                 "finish_reason": fin.finish_reason if fin else "length"}],
                     "usage": _usage(fin.prompt_len, len(fin.output_token_ids)) if fin else None,
                 }
                 yield f"data: {json.dumps(final)}\n\n"

@@ -371,6 +371,16 @@ class EngineHandle:
         self.adapter_registry: Dict[str, Any] = {}
         self._adapter_events: Dict[str, threading.Event] = {}
         self._adapter_errors: Dict[str, Optional[str]] = {}
+        # Connection.send splits payloads >16 KiB into multiple OS writes;
+        # submit/abort run on the event loop while adapter ops run in executor
+        # threads, so every send must hold this lock or the command pipe can
+        # interleave and corrupt (kills the worker under adapter rollout +
+        # live traffic).
+        self._send_lock = threading.Lock()
+
+    def _send(self, msg) -> None:
+        with self._send_lock:
+            self.conn.send(msg)
 
     @property
     def tp_size(self) -> int:
@@ -457,10 +467,10 @@ class EngineHandle:
 
     # -- command API --------------------------------------------------------
     def submit(self, req: PoolRequest) -> None:
-        self.conn.send(("add", req))
+        self._send(("add", req))
 
     def abort(self, request_id: str) -> None:
-        self.conn.send(("abort", request_id))
+        self._send(("abort", request_id))
 
     def load_adapter(
         self, name: str, source: Any = None, rank: Optional[int] = None,
@@ -473,7 +483,7 @@ class EngineHandle:
         self.adapter_registry[name] = (source, rank)
         ev = threading.Event()
         self._adapter_events[name] = ev
-        self.conn.send(("load_adapter", name, source, rank))
+        self._send(("load_adapter", name, source, rank))
         if not wait:
             return None
         if not ev.wait(timeout_s):
@@ -485,17 +495,17 @@ class EngineHandle:
         self.adapter_registry.pop(name, None)
         ev = threading.Event()
         self._adapter_events[name] = ev
-        self.conn.send(("unload_adapter", name))
+        self._send(("unload_adapter", name))
         if wait:
             ev.wait(timeout_s)
 
     def drain(self, on: bool = True) -> None:
-        self.conn.send(("drain", on))
+        self._send(("drain", on))
 
     def stop(self) -> None:
         if self.conn is not None:
             try:
-                self.conn.send(("stop",))
+                self._send(("stop",))
             except (BrokenPipeError, OSError):
                 pass
         for p in self.procs:

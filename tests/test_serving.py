@@ -96,6 +96,30 @@ def test_streaming_sse(client):
     assert len(events) == 5
 
 
+def test_streaming_holdback_with_stop(client):
+    """With a stop string present the stream holds back len(stop) tokens
+    until the stop check clears them; nothing is lost when no stop matches,
+    and the final chunk carries the engine's real finish_reason (not a
+    hardcoded 'length')."""
+    with client.stream(
+        "POST",
+        "/v1/completions",
+        json={"model": "my-model", "prompt": "hold back", "max_tokens": 4,
+              "stream": True, "stop": "XYZ"},
+    ) as r:
+        assert r.status_code == 200
+        events = [ln for ln in r.iter_lines() if ln.startswith("data: ")]
+    assert events[-1] == "data: [DONE]"
+    import json as _json
+
+    chunks = [_json.loads(e[len("data: "):]) for e in events[:-1]]
+    token_chunks = [c for c in chunks if c["choices"][0].get("token") is not None]
+    assert len(token_chunks) == 4  # all 4 tokens delivered despite holdback
+    final = chunks[-1]
+    assert final["choices"][0]["finish_reason"] == "length"
+    assert final["usage"]["completion_tokens"] == 4
+
+
 def test_traffic_split_resolves_to_adapter(client, pool):
     """split-model routes 100% to adapter-a (reference weighted draw +
     body-rewrite semantics, request.go:47-70)."""
