@@ -74,6 +74,18 @@ class DecodeGraphRunner:
         self._pool = None
         self.with_lora = False
 
+    def _graph_nsplits(self, B: int) -> int:
+        """Bucket-static flash-decode split factor: replay can't branch on
+        live context lengths, so pick from the bucket size + the engine's max
+        context (empty split ranges merge as no-ops — bit-exact either way,
+        profiles/r01_flash_decode_split.txt)."""
+        from .. import ops
+
+        attn = self.runner.model.layers[0].self_attn
+        return ops.decode_nsplits(
+            B, attn.num_kv_heads, self.max_blocks * self.block_size
+        )
+
     def _meta(self, B: int) -> BatchMeta:
         return BatchMeta(
             num_prefill_tokens=0,
@@ -82,6 +94,7 @@ class DecodeGraphRunner:
             decode_block_tables=self.block_tables[:B],
             decode_context_lens=self.context_lens[:B],
             lora_slot_per_token=self.lora_slots[:B] if self.with_lora else None,
+            decode_nsplits=self._graph_nsplits(B),
         )
 
     def _forward(self, B: int) -> torch.Tensor:
@@ -92,6 +105,20 @@ class DecodeGraphRunner:
 
     def capture(self, with_lora: bool) -> None:
         self.with_lora = with_lora
+        # Preallocate the flash-decode split scratch at the max any bucket
+        # needs so every captured graph bakes one stable pointer (the ops
+        # cache grows monotonically; growth during capture would leave
+        # earlier graphs pointing at freed memory).
+        from .. import ops
+
+        attn = self.runner.model.layers[0].self_attn
+        hq, d = attn.num_heads, attn.head_dim
+        max_scratch = max(
+            (B * hq * self._graph_nsplits(B) * (d + 2) for B in self.sizes),
+            default=0,
+        )
+        if max_scratch > 0 and self.device.type == "cuda":
+            ops.split_scratch(self.device, max_scratch)
         # seed static buffers with safe values (padding semantics)
         self.block_tables.fill_(0)
         self.block_tables[:, 0] = self.scratch_block

@@ -42,9 +42,13 @@ class BatchMeta:
 
     # Multi-LoRA (SGMV): per-token adapter slot index, -1 = base model only.
     lora_slot_per_token: Optional[torch.Tensor] = None  # [T] int32
-    # host-side max decode context this step (drives the round-2 flash-decode
-    # split heuristic without a device sync)
+    # host-side max decode context this step (drives the flash-decode split
+    # heuristic without a device sync)
     max_decode_ctx: int = 0
+    # pinned context-split factor for the decode kernel (0 = let the ops
+    # heuristic decide from max_decode_ctx; the hipGraph path captures a
+    # bucket-static value here since replay cannot branch on context)
+    decode_nsplits: int = 0
     # Prefill LoRA segment tiles (64-token runs sharing one adapter slot) for
     # the MFMA SGMV kernels: token start, length (<=64), adapter slot.
     lora_tile_t0: Optional[torch.Tensor] = None    # [ntiles] int32

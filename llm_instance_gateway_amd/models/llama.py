@@ -103,6 +103,8 @@ class Attention(nn.Module):
                 meta.decode_block_tables,
                 meta.decode_context_lens,
                 self.scale,
+                max_ctx=meta.max_decode_ctx,
+                nsplits=meta.decode_nsplits,
             )
         attn_flat = out.view(-1, self.num_heads * self.head_dim)
         o = self.o_proj(attn_flat)

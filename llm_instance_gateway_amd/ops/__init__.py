@@ -107,6 +107,46 @@ def reshape_and_cache(
     torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
 
 
+# Flash-decoding context split (ROADMAP r01 #1, measured in
+# profiles/r01_flash_decode_split.txt: 13.7x at N=4/ctx=16k, neutral at
+# chip-filling batch).  The single-pass kernel launches N*Hkv workgroups;
+# when that is far below the chip's 256 CUs and the context is long, split
+# the context across `ns` workgroups + a combine pass (bit-exact merge).
+_NUM_CUS = 256
+_SPLIT_MIN_CTX = 1024      # short contexts: single-pass latency already fine
+_SPLIT_MIN_PER_SPLIT = 256  # keep each split's K/V walk worth a workgroup
+_SPLIT_MAX = 32
+_SPLIT_SCRATCH: dict = {}   # device -> fp32 scratch tensor
+# Superseded scratches stay referenced: a captured hipGraph bakes the pointer
+# it saw, so freeing an old scratch after a grow would let the allocator hand
+# that memory to live tensors the graph then scribbles over.
+_SPLIT_SCRATCH_RETIRED: list = []
+
+
+def decode_nsplits(n_seqs: int, n_kv_heads: int, max_ctx: int) -> int:
+    """Context-split factor for the paged decode kernel (1 = single-pass)."""
+    wg = n_seqs * n_kv_heads
+    target = 2 * _NUM_CUS
+    if wg == 0 or wg >= target or max_ctx < _SPLIT_MIN_CTX:
+        return 1
+    ns = min(_SPLIT_MAX, (target + wg - 1) // wg)
+    ns = min(ns, max(1, max_ctx // _SPLIT_MIN_PER_SPLIT))
+    return max(1, ns)
+
+
+def split_scratch(device, numel: int) -> torch.Tensor:
+    """fp32 partials scratch [N, Hq, ns, D+2] (flat).  Cached per device and
+    grown monotonically; graph capture preallocates its max so every captured
+    graph keeps one stable pointer."""
+    t = _SPLIT_SCRATCH.get(device)
+    if t is None or t.numel() < numel:
+        if t is not None:
+            _SPLIT_SCRATCH_RETIRED.append(t)
+        t = torch.empty(numel, dtype=torch.float32, device=device)
+        _SPLIT_SCRATCH[device] = t
+    return t
+
+
 def paged_attention_decode(
     q: torch.Tensor,
     k_cache: torch.Tensor,
@@ -114,12 +154,28 @@ def paged_attention_decode(
     block_tables: torch.Tensor,
     context_lens: torch.Tensor,
     scale: float,
+    max_ctx: int = 0,
+    nsplits: int = 0,
 ) -> torch.Tensor:
+    """`max_ctx` is a host-side hint (no device sync) driving the split
+    heuristic; `nsplits` > 0 pins the factor (the hipGraph path captures a
+    bucket-static value)."""
     if _use_native(q):
         out = torch.empty_like(q)
-        _EXT.paged_attention_decode(
-            out, q, k_cache, v_cache, block_tables, context_lens, scale, _stream()
-        )
+        N, Hq, D = q.shape
+        Hkv = k_cache.shape[1]
+        ns = nsplits if nsplits > 0 else decode_nsplits(N, Hkv, max_ctx)
+        if ns > 1:
+            scratch = split_scratch(q.device, N * Hq * ns * (D + 2))
+            _EXT.paged_attention_decode_split(
+                out, q, k_cache, v_cache, block_tables, context_lens,
+                scratch, scale, ns, _stream()
+            )
+        else:
+            _EXT.paged_attention_decode(
+                out, q, k_cache, v_cache, block_tables, context_lens, scale,
+                _stream()
+            )
         return out
     return torch_ref.paged_attention_decode(
         q, k_cache, v_cache, block_tables, context_lens, scale

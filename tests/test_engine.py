@@ -222,3 +222,21 @@ def test_critical_requests_admitted_first():
     ]
     eng.generate_all()
     assert crit.first_scheduled_time < normal.first_scheduled_time
+
+
+def test_decode_nsplits_heuristic():
+    """Flash-decode split factor: off at chip-filling batch or short context,
+    on (bounded) for small-batch long-context shapes."""
+    from llm_instance_gateway_amd import ops
+
+    # chip-filling: 256 seqs x 8 kv-heads = 2048 workgroups >> 512 target
+    assert ops.decode_nsplits(256, 8, 16384) == 1
+    # short context: single-pass regardless of batch
+    assert ops.decode_nsplits(4, 8, 512) == 1
+    # small batch, long context: split toward ~2*CUs workgroups
+    ns = ops.decode_nsplits(4, 8, 16384)
+    assert 8 <= ns <= 32
+    # per-split work floor: never split below ~256 ctx tokens per split
+    assert ops.decode_nsplits(2, 8, 1024) <= 4
+    # degenerate
+    assert ops.decode_nsplits(0, 8, 16384) == 1

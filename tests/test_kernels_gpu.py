@@ -267,6 +267,26 @@ def test_sgmv_tiled_matches_ref(R, O):
     close(y, y_ref.to(torch.bfloat16))
 
 
+@pytest.mark.parametrize("N,ctx", [(4, 2048), (2, 1536)])
+def test_decode_split_default_dispatch(N, ctx):
+    """ops.paged_attention_decode with the max_ctx hint routes small-batch
+    long-context decodes to the split kernel (heuristic: N*Hkv workgroups
+    << 2*CUs) and matches the single-pass result."""
+    HKV, HQ, D, BS = 8, 32, 128, 16
+    assert ops.decode_nsplits(N, HKV, ctx) > 1  # heuristic fires for this shape
+    kc, vc, bt, cl = _rand_cache_and_tables(N, HKV, D, BS, ctx, seed=21)
+    q = torch.randn(N, HQ, D, dtype=torch.bfloat16, device=DEV) * 0.3
+    scale = D ** -0.5
+    single = ops.paged_attention_decode(q, kc, vc, bt, cl, scale)  # no hint
+    split = ops.paged_attention_decode(q, kc, vc, bt, cl, scale, max_ctx=ctx)
+    close(split, single)
+    # pinned factor (the graph path's bucket-static value)
+    pinned = ops.paged_attention_decode(
+        q, kc, vc, bt, cl, scale, nsplits=8
+    )
+    close(pinned, single)
+
+
 @pytest.mark.parametrize("N,ctx,ns", [(4, 300, 2), (16, 1024, 4), (3, 7, 3), (8, 2048, 16)])
 def test_paged_attention_decode_split_matches(N, ctx, ns):
     """Flash-decoding context split + combine == single-pass kernel
