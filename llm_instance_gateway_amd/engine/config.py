@@ -39,6 +39,13 @@ class EngineConfig:
     engine_id: str = "engine-0"
     gpu_id: int = 0
     enforce_eager: bool = False  # False: decode-only steps replay hipGraphs
+    # Pad mixed (eager) batches to a multiple of this many tokens so the
+    # prefill GEMMs see a handful of stable, tile-friendly M shapes instead
+    # of a new M every step (hipBLASLt heuristic picks at variable M ran at
+    # ~17% of bf16 peak — profiles/r01_steady_state_breakdown.md).  Pad rows
+    # write K/V to the scratch block and are excluded from attention,
+    # LoRA, and sampling.  0 disables; GPU only; dense models only.
+    pad_batch_multiple: int = 256
 
     def __post_init__(self) -> None:
         if self.block_size <= 0:

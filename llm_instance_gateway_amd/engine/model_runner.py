@@ -152,6 +152,25 @@ class ModelRunner:
             sample_seqs.append(seq)
             sample_idx.append(len(input_ids) - 1)
 
+        # GEMM M-bucketing: pad the flat batch tail so every projection GEMM
+        # sees M % pad_batch_multiple == 0 (few stable shapes -> hipBLASLt
+        # algo reuse + tile-friendly M).  Pad rows: token 0 at position 0,
+        # K/V scattered to the scratch block (same slot — racy garbage is
+        # fine there), outside both attention partitions and the sample set.
+        num_pad_tokens = 0
+        padm = self.config.pad_batch_multiple
+        if (
+            padm > 0
+            and self.device.type == "cuda"
+            and not self.config.model.is_moe
+            and len(input_ids) % padm != 0
+        ):
+            num_pad_tokens = padm - len(input_ids) % padm
+            scratch_slot = self.config.num_kv_blocks * bs
+            input_ids.extend([0] * num_pad_tokens)
+            positions.extend([0] * num_pad_tokens)
+            slots.extend([scratch_slot] * num_pad_tokens)
+
         dev = self.device
 
         def pad_tables(bts: List[List[int]]) -> Optional[torch.Tensor]:
@@ -240,6 +259,8 @@ class ModelRunner:
                         tslots.append(s_id)
             for ss in decodes:
                 slots.append(mgr.slot_of(ss.seq.lora_name))
+            if num_pad_tokens:
+                slots.extend([-1] * num_pad_tokens)  # pad rows: base model only
             meta.lora_slot_per_token = torch.tensor(slots, dtype=torch.int32).to(
                 dev, non_blocking=True
             )

@@ -96,8 +96,10 @@ class Attention(nn.Module):
                 tile_qstart=meta.prefill_tile_qstart,
             )
         if nd > 0:
-            out[np_:] = ops.paged_attention_decode(
-                qh[np_:],
+            # slice end is explicit: the flat batch may carry GEMM-bucketing
+            # pad rows after the decode partition
+            out[np_:np_ + nd] = ops.paged_attention_decode(
+                qh[np_:np_ + nd],
                 k_cache,
                 v_cache,
                 meta.decode_block_tables,
