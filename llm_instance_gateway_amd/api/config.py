@@ -104,6 +104,7 @@ def parse_config(raw: Dict[str, Any]) -> GatewayConfig:
             engine_addresses=[str(a) for a in p.get("engine_addresses", [])],
             base_model=str(p.get("base_model", "")),
             tensor_parallel=int(p.get("tensor_parallel", 1)),
+            devices=[str(d) for d in p.get("devices", [])],
         )
 
     for m in raw.get("models", []) or []:
@@ -184,6 +185,56 @@ def reconcile_models(cfg: GatewayConfig, datastore) -> None:
     for existing in datastore.all_models():
         if existing.model_name not in wanted:
             datastore.delete_model(existing.model_name)
+
+
+def desired_engine_specs(
+    cfg: GatewayConfig,
+    default_model: str = "",
+    engine_kwargs: Optional[Dict[str, Any]] = None,
+    lora_rank: int = 16,
+) -> Optional[List[Any]]:
+    """Derive the desired engine set from the pool config (the reference's
+    EndpointSlice -> Pod-set mapping, `endpointslice_reconciler.go:50-79`):
+    ``pool.devices`` (explicit device names) or ``pool.gpu_ids`` (cuda:<id>),
+    chunked into TP groups of ``tensor_parallel`` devices.  Returns None when
+    the config does not pin membership (empty devices AND gpu_ids) — then the
+    reconciler leaves the running set alone."""
+    from ..serving.pool import EngineSpec
+
+    p = cfg.pool
+    devs = list(p.devices) if p.devices else [f"cuda:{g}" for g in p.gpu_ids]
+    if not devs:
+        return None
+    tp = max(1, p.tensor_parallel)
+    if len(devs) % tp != 0:
+        raise ConfigError(
+            f"pool lists {len(devs)} devices, not a multiple of tensor_parallel={tp}"
+        )
+    model = p.base_model or default_model
+    specs = []
+    for i in range(0, len(devs), tp):
+        specs.append(
+            EngineSpec(
+                name=f"engine-{i // tp}",
+                model=model,
+                devices=devs[i : i + tp],
+                engine_kwargs=dict(engine_kwargs or {}),
+                lora_rank=lora_rank,
+                pool_ref=p.name,
+            )
+        )
+    return specs
+
+
+def reconcile_pool_membership(cfg: GatewayConfig, pool) -> Tuple[List[str], List[str]]:
+    """Grow/shrink the live engine set to match the config (scale-out spawns
+    workers, scale-in drains then stops).  Returns (added, removed)."""
+    specs = desired_engine_specs(
+        cfg, pool.model_name, pool.engine_kwargs, pool.lora_rank
+    )
+    if specs is None:
+        return [], []
+    return pool.reconcile_membership(specs)
 
 
 def reconcile_adapters(cfg: GatewayConfig, pool) -> Tuple[Set[str], Set[str], Set[str]]:
@@ -315,6 +366,15 @@ class ConfigReconciler:
         if self.datastore is not None:
             reconcile_models(cfg, self.datastore)
         if self.pool is not None:
+            # membership first: engines added here pick up the adapter set in
+            # the same pass (the reference's EndpointSlice reconcile runs
+            # independently of the sidecar; collapsing them keeps ordering)
+            try:
+                added, removed = reconcile_pool_membership(cfg, self.pool)
+                if added or removed:
+                    log.info("pool membership: +%s -%s", added, removed)
+            except (ConfigError, RuntimeError) as e:
+                log.error("pool membership reconcile failed: %s", e)
             reconcile_adapters(cfg, self.pool)
         for act in self.remote_actuators:
             reconcile_adapters_remote(cfg, act)

@@ -100,6 +100,9 @@ class InferencePool:
     # Base model served by the pool's engines and its parallelism degree.
     base_model: str = ""
     tensor_parallel: int = 1
+    # Explicit device names (overrides gpu_ids; lets CPU deployments/tests
+    # drive runtime membership reconciliation the same way).
+    devices: List[str] = field(default_factory=list)
 
 
 def random_weighted_draw(model: InferenceModel, seed: int = 0) -> str:

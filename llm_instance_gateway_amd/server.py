@@ -37,8 +37,8 @@ def build_pool(args, cfg) -> "EnginePool":
     model = args.model or (cfg.pool.base_model if cfg else "") or "llama3-8b"
     if args.devices:
         devices: List[str] = args.devices.split(",")
-    elif cfg and cfg.pool.gpu_ids:
-        devices = [f"cuda:{g}" for g in cfg.pool.gpu_ids]
+    elif cfg and (cfg.pool.devices or cfg.pool.gpu_ids):
+        devices = cfg.pool.devices or [f"cuda:{g}" for g in cfg.pool.gpu_ids]
     elif torch.cuda.is_available():
         devices = [f"cuda:{i}" for i in range(torch.cuda.device_count())]
     else:
@@ -50,7 +50,16 @@ def build_pool(args, cfg) -> "EnginePool":
             num_kv_blocks=256, max_num_seqs=16,
             max_num_batched_tokens=512, max_model_len=512,
         )
-    pool = EnginePool(model, devices=devices, engine_kwargs=engine_kwargs)
+    tp = cfg.pool.tensor_parallel if cfg else 1
+    if cfg and tp > 1 and not args.devices:
+        # TP pools: chunk devices into TP groups, one engine per group —
+        # the same shape the config reconciler derives at runtime
+        from .api.config import desired_engine_specs
+
+        specs = desired_engine_specs(cfg, model, engine_kwargs)
+        pool = EnginePool(model, engine_kwargs=engine_kwargs, specs=specs)
+    else:
+        pool = EnginePool(model, devices=devices, engine_kwargs=engine_kwargs)
     if cfg:
         tree = build_default_tree(
             kv_cache_threshold=cfg.kv_cache_threshold,

@@ -565,11 +565,16 @@ class EnginePool:
         datastore: Optional[Datastore] = None,
         scheduler: Optional[Scheduler] = None,
         specs: Optional[Sequence["EngineSpec"]] = None,
+        max_engines: int = 0,
     ):
         """Homogeneous pool: ``EnginePool(model, devices=[...])`` — one TP=1
         engine per device (the reference's replica fleet).  Mixed pool:
         ``EnginePool(specs=[EngineSpec(...), ...])`` — arbitrary per-engine
-        model / device-set / TP degree (BASELINE config #4)."""
+        model / device-set / TP degree (BASELINE config #4).
+
+        ``max_engines`` bounds LIFETIME membership capacity (shared gauge
+        slots are pre-sized so the pool can grow at runtime — the reference's
+        EndpointSlice add/prune, `endpointslice_reconciler.go:50-79`)."""
         if specs is None:
             specs = [
                 EngineSpec(
@@ -583,35 +588,54 @@ class EnginePool:
             ]
         self.model_name = model_name or (specs[0].model if specs else "")
         self.engine_kwargs = dict(engine_kwargs or {})
+        self.lora_rank = lora_rank
+        self.max_engines = max(max_engines, 16, 2 * len(specs))
         self.shared = mp.get_context("spawn").Array(
-            "d", SLOTS_PER_ENGINE * len(specs), lock=False
+            "d", SLOTS_PER_ENGINE * self.max_engines, lock=False
         )
+        self._free_slots: List[int] = [
+            SLOTS_PER_ENGINE * i for i in range(self.max_engines - 1, -1, -1)
+        ]
         self.datastore = datastore or Datastore()
         self.handles: Dict[str, EngineHandle] = {}
-        for i, spec in enumerate(specs):
-            dev0 = spec.devices[0]
-            eng = Engine(
-                name=spec.name,
-                gpu_id=i if dev0.startswith("cuda") else -1,
-                model=spec.model,
-                pool=spec.pool_ref,
-            )
-            self.handles[eng.name] = EngineHandle(
-                self,
-                eng,
-                slot_base=SLOTS_PER_ENGINE * i,
-                model_name=spec.model,
-                devices=spec.devices,
-                engine_kwargs=dict(spec.engine_kwargs),
-                lora_rank=spec.lora_rank,
-            )
-            self.datastore.add_engine(eng)
+        for spec in specs:
+            self._make_handle(spec)
         self.scheduler = scheduler or Scheduler(self)
         self._tickets: Dict[str, RequestTicket] = {}
         self._tickets_lock = threading.Lock()
         self.on_finished: Optional[Callable[[FinishedRequest], None]] = None
         self._watchdog: Optional[threading.Thread] = None
         self._stop = threading.Event()
+
+    def _make_handle(self, spec: "EngineSpec") -> EngineHandle:
+        """Allocate a gauge slot + handle + datastore entry (not started)."""
+        if spec.name in self.handles:
+            raise ValueError(f"engine {spec.name} already in the pool")
+        if not self._free_slots:
+            raise RuntimeError(f"pool at max_engines={self.max_engines}")
+        slot_base = self._free_slots.pop()
+        dev0 = spec.devices[0]
+        gpu_id = int(dev0.split(":")[1]) if dev0.startswith("cuda:") else (
+            0 if dev0 == "cuda" else -1
+        )
+        eng = Engine(
+            name=spec.name,
+            gpu_id=gpu_id,
+            model=spec.model,
+            pool=spec.pool_ref,
+        )
+        h = EngineHandle(
+            self,
+            eng,
+            slot_base=slot_base,
+            model_name=spec.model,
+            devices=spec.devices,
+            engine_kwargs=dict(spec.engine_kwargs),
+            lora_rank=spec.lora_rank,
+        )
+        self.handles[eng.name] = h
+        self.datastore.add_engine(eng)
+        return h
 
     # -- lifecycle ----------------------------------------------------------
     def start(self, watchdog: bool = True) -> None:
@@ -633,7 +657,7 @@ class EnginePool:
 
     def stop(self) -> None:
         self._stop.set()
-        for h in self.handles.values():
+        for h in list(self.handles.values()):
             h.stop()
 
     def _watchdog_loop(self, interval_s: float = 1.0) -> None:
@@ -641,7 +665,7 @@ class EnginePool:
         the engine from the routing set — reference Ready-condition semantics
         (`endpointslice_reconciler.go:107-111`)."""
         while not self._stop.wait(interval_s):
-            for h in self.handles.values():
+            for h in list(self.handles.values()):  # membership may change live
                 stale = h.heartbeat_age_s() > HEARTBEAT_STALE_S
                 dead = bool(h.procs) and any(not p.is_alive() for p in h.procs)
                 if (stale or dead) and h.healthy:
@@ -675,6 +699,66 @@ class EnginePool:
             h.load_adapter(aname, source, rank, wait=False)
         self.datastore.add_engine(h.engine)
 
+    # -- runtime membership (reference EndpointSlice add/prune,
+    #    endpointslice_reconciler.go:50-79) ----------------------------------
+    def add_engine_live(self, spec: "EngineSpec") -> EngineHandle:
+        """Scale-out: spawn a new engine worker into the running pool; it
+        joins the routing set once its first heartbeat lands."""
+        h = self._make_handle(spec)
+        try:
+            h.start()
+        except Exception:
+            self.datastore.remove_engine(spec.name)
+            del self.handles[spec.name]
+            self._free_slots.append(h.slot_base)
+            raise
+        log.info("engine %s added to pool (device %s)", spec.name, spec.devices)
+        return h
+
+    def remove_engine_live(
+        self, name: str, drain: bool = True, drain_timeout_s: float = 60.0
+    ) -> None:
+        """Scale-in: drop the engine from the routing set immediately, drain
+        in-flight work (bounded), then stop the worker and free its slot."""
+        h = self.handles.get(name)
+        if h is None:
+            return
+        self.datastore.remove_engine(name)
+        if drain and h.alive():
+            h.drain(True)
+            deadline = time.monotonic() + drain_timeout_s
+            while time.monotonic() < deadline:
+                with self._tickets_lock:
+                    inflight = any(
+                        t.engine_name == name for t in self._tickets.values()
+                    )
+                if not inflight:
+                    break
+                time.sleep(0.02)
+        h.stop()
+        self._fail_inflight(name, "engine removed from pool")
+        del self.handles[name]
+        self._free_slots.append(h.slot_base)
+        log.info("engine %s removed from pool", name)
+
+    def reconcile_membership(
+        self, specs: Sequence["EngineSpec"], drain_timeout_s: float = 60.0
+    ) -> tuple:
+        """Make the live engine set match ``specs`` by name: start engines
+        that are missing, drain-and-remove engines no longer desired.
+        Returns (added_names, removed_names)."""
+        desired = {s.name: s for s in specs}
+        current = set(self.handles)
+        added, removed = [], []
+        for name in sorted(current - set(desired)):
+            self.remove_engine_live(name, drain=True, drain_timeout_s=drain_timeout_s)
+            removed.append(name)
+        for name, spec in desired.items():
+            if name not in self.handles:
+                self.add_engine_live(spec)
+                added.append(name)
+        return added, removed
+
     # -- EngineMetricsProvider protocol -------------------------------------
     def all_engine_metrics(self) -> List[EngineState]:
         out = []
@@ -691,7 +775,11 @@ class EnginePool:
         return self.scheduler.schedule(req)
 
     def submit(self, engine_name: str, req: PoolRequest) -> RequestTicket:
-        h = self.handles[engine_name]
+        h = self.handles.get(engine_name)
+        if h is None:
+            # engine removed between route and dispatch (scale-in race):
+            # surface as retryable, same as the died-mid-dispatch path below
+            raise SchedulingError(f"engine {engine_name} left the pool")
         ticket = RequestTicket(req.request_id, engine_name, req.stream)
         with self._tickets_lock:
             self._tickets[req.request_id] = ticket
