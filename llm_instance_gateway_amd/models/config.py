@@ -81,6 +81,22 @@ MIXTRAL_8X7B = ModelConfig(
     num_experts_per_tok=2,
 )
 
+# 70B-shaped tiny: keeps every ratio TP=4 sharding depends on (64 q heads /
+# 8 kv heads GQA grouping, 3.5x intermediate, separate lm_head) at CPU-test
+# dims — the 4-rank gloo token-identity test for BASELINE config #4 runs on
+# this shape.
+TINY_70B = ModelConfig(
+    name="tiny-70b",
+    vocab_size=512,
+    hidden_size=128,
+    num_layers=2,
+    num_heads=64,
+    num_kv_heads=8,
+    intermediate_size=448,
+    head_dim=16,
+    max_position_embeddings=512,
+)
+
 # Tiny shapes for CPU tests and smoke runs.
 TINY = ModelConfig(
     name="tiny",
@@ -111,6 +127,7 @@ TINY_MOE = ModelConfig(
 PRESETS = {
     "llama3-8b": LLAMA3_8B,
     "llama3-70b": LLAMA3_70B,
+    "tiny-70b": TINY_70B,
     "mixtral-8x7b": MIXTRAL_8X7B,
     "tiny": TINY,
     "tiny-moe": TINY_MOE,

@@ -69,10 +69,10 @@ def _shard_llama(full, sharded, rank: int, tp: int) -> None:
             )
 
 
-def _tp_worker(rank: int, world: int, q):
+def _tp_worker(rank: int, world: int, q, model: str = "tiny", port: int = PORT):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
-        os.environ["MASTER_PORT"] = str(PORT)
+        os.environ["MASTER_PORT"] = str(port)
         from llm_instance_gateway_amd.engine import EngineConfig, LLMEngine, SamplingParams
         from llm_instance_gateway_amd.parallel import TPGroup, init_distributed
 
@@ -82,11 +82,13 @@ def _tp_worker(rank: int, world: int, q):
         # collective sanity
         t = torch.ones(4) * (rank + 1)
         tp.all_reduce(t)
-        assert torch.allclose(t, torch.full((4,), 3.0)), t
+        tot = float(world * (world + 1) // 2)
+        assert torch.allclose(t, torch.full((4,), tot)), t
         g = tp.all_gather(torch.full((2,), float(rank)), dim=0)
-        assert torch.allclose(g, torch.tensor([0.0, 0.0, 1.0, 1.0])), g
+        exp = torch.tensor([float(r) for r in range(world) for _ in range(2)])
+        assert torch.allclose(g, exp), g
 
-        mc = get_model_config("tiny")
+        mc = get_model_config(model)
         ekw = dict(
             num_kv_blocks=64,
             max_num_seqs=4,
@@ -182,17 +184,31 @@ def _tp_worker(rank: int, world: int, q):
         q.put((rank, f"{e}\n{traceback.format_exc()}"))
 
 
-def test_tp2_token_identical_generation():
+def _run_tp(world: int, model: str, port: int) -> None:
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_tp_worker, args=(r, 2, q)) for r in range(2)]
+    procs = [
+        ctx.Process(target=_tp_worker, args=(r, world, q, model, port))
+        for r in range(world)
+    ]
     for p in procs:
         p.start()
-    results = [q.get(timeout=300) for _ in range(2)]
+    results = [q.get(timeout=300) for _ in range(world)]
     for p in procs:
         p.join(timeout=60)
     for rank, status in results:
         assert status == "ok", f"rank {rank}: {status}"
+
+
+def test_tp2_token_identical_generation():
+    _run_tp(2, "tiny", PORT)
+
+
+def test_tp4_token_identical_generation_70b_shape():
+    """BASELINE config #4 readiness: TP=4 over gloo, 70B-shaped layer
+    (64 q heads / 8 kv heads GQA, 3.5x intermediate), token-identical to the
+    unsharded engine, base + sharded-adapter paths."""
+    _run_tp(4, "tiny-70b", PORT + 7)
 
 
 def test_shard_sizes():
