@@ -100,11 +100,9 @@ class ModelRunner:
         )
         self.adapter_manager = None
         self.graph_runner = None
-        # MoE dispatch is token-dependent; a captured decode graph would
-        # freeze the capture batch's expert routing -> eager for MoE.
-        self._use_graphs = (
-            self.device.type == "cuda" and not config.enforce_eager and not mc.is_moe
-        )
+        # MoE decode is graph-safe too: decode-only batches use the dense
+        # expert dispatch (static shapes, no host sync — models/mixtral.py).
+        self._use_graphs = self.device.type == "cuda" and not config.enforce_eager
 
     # -- batch construction ---------------------------------------------------
     def _build_batch(self, out: SchedulerOutput):

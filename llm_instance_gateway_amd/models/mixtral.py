@@ -17,9 +17,12 @@ MI355X design notes:
     Padding rows read a zero row and scatter back with weight 0 (bit-exact,
     no token dropping; verified against the per-token reference in
     tests/test_mixtral.py).
-  * MoE engines run eager (no hipGraph decode capture): expert routing is
-    token-dependent, so a captured graph would freeze the dispatch of the
-    capture batch (`engine/model_runner.py` gates on `is_moe`).
+  * Decode-only batches use a DENSE expert dispatch (every expert computes
+    every token; unselected experts gated by weight 0): static shapes and no
+    host sync make decode steps hipGraph-capturable, and decode MoE is bound
+    by streaming all expert weights (2.8 GB/layer for 8x7B) regardless of
+    routing, so the extra rows are nearly free.  Prefill/mixed batches —
+    compute-bound — use the capacity-routed bmm dispatch.
 
 LoRA: adapters attach to the attention projections only (qkv / o); routed
 expert weights are not adapter targets — matching what the reference fleet
@@ -74,6 +77,30 @@ class MoEBlock(nn.Module):
         weights, experts = torch.topk(router_logits, self.top_k, dim=-1)
         # Mixtral normalizes softmax over the selected k only
         weights = torch.softmax(weights.float(), dim=-1).to(x.dtype)  # [T, k]
+
+        # Decode-only batches take the DENSE dispatch: every expert computes
+        # every token, unselected experts gated by weight 0.  Static shapes,
+        # no host sync -> hipGraph-capturable; and decode MoE is bound by
+        # streaming ALL expert weights regardless of routing (2.8 GB/layer
+        # for 8x7B), so the extra rows are nearly free.  Prefill/mixed
+        # batches (compute-bound) keep the capacity-routed path below.
+        if meta is not None and meta.num_prefill_tokens == 0:
+            wdense = torch.zeros(
+                T, self.num_experts, dtype=x.dtype, device=x.device
+            )
+            wdense.scatter_(1, experts, weights)
+            E = self.num_experts
+            xe = x.unsqueeze(0).expand(E, T, h)
+            act = ops.silu_and_mul(torch.bmm(xe, self.w13).view(E * T, -1))
+            ye = torch.bmm(act.view(E, T, -1), self.w2)  # [E, T, h]
+            out = (
+                (ye.float() * wdense.t().unsqueeze(-1).float())
+                .sum(dim=0)
+                .to(x.dtype)
+            )
+            if self.tp is not None:
+                out = self.tp.all_reduce(out)
+            return out
 
         flat_experts = experts.reshape(-1)  # [T*k]
         flat_tokens = (

@@ -67,3 +67,31 @@ def test_moe_batched_equals_sequential_greedy():
     done = sorted(eng2.generate_all(), key=lambda s: s.prompt_len)
     batch_outs = [s.output_token_ids for s in done]
     assert batch_outs == seq_outs
+
+
+def test_moe_dense_decode_path_matches_routed():
+    """The graph-safe dense dispatch (decode-only batches) == the routed
+    capacity path == the per-token reference."""
+    from llm_instance_gateway_amd.engine.metadata import BatchMeta
+
+    mc = get_model_config("tiny-moe")
+    torch.manual_seed(21)
+    blk = MoEBlock(mc, device="cpu", dtype=torch.float32)
+    T = 9
+    x = torch.randn(T, mc.hidden_size)
+    routed = blk(x)  # meta None -> routed path
+    meta = BatchMeta(
+        num_prefill_tokens=0,
+        num_decode_tokens=T,
+        slot_mapping=torch.zeros(T, dtype=torch.int64),
+    )
+    dense = blk(x, meta)
+    assert torch.allclose(dense, routed, atol=1e-5), (dense - routed).abs().max()
+    # mixed batch still takes the routed path (prefill tokens present)
+    meta_mixed = BatchMeta(
+        num_prefill_tokens=4,
+        num_decode_tokens=T - 4,
+        slot_mapping=torch.zeros(T, dtype=torch.int64),
+    )
+    mixed = blk(x, meta_mixed)
+    assert torch.allclose(mixed, routed, atol=1e-5)
