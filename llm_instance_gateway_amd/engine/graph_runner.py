@@ -53,6 +53,11 @@ class DecodeGraphRunner:
         )
         self.context_lens = torch.ones(B, dtype=torch.int32, device=dev)
         self.lora_slots = torch.full((B,), -1, dtype=torch.int32, device=dev)
+        # slot-grouped SGMV unit buffers (U == B, empty units padded len=0)
+        self.lora_order = torch.zeros(B, dtype=torch.int32, device=dev)
+        self.lora_ut0 = torch.zeros(B, dtype=torch.int32, device=dev)
+        self.lora_ulen = torch.zeros(B, dtype=torch.int32, device=dev)
+        self.lora_uslot = torch.full((B,), -1, dtype=torch.int32, device=dev)
         # pinned host staging (one copy per tensor per step)
         self.h_ids = torch.zeros(B, dtype=torch.int64, pin_memory=True)
         self.h_pos = torch.zeros(B, dtype=torch.int64, pin_memory=True)
@@ -60,6 +65,10 @@ class DecodeGraphRunner:
         self.h_bt = torch.zeros(B, self.max_blocks, dtype=torch.int32, pin_memory=True)
         self.h_ctx = torch.ones(B, dtype=torch.int32, pin_memory=True)
         self.h_lora = torch.full((B,), -1, dtype=torch.int32, pin_memory=True)
+        self.h_order = torch.zeros(B, dtype=torch.int32, pin_memory=True)
+        self.h_ut0 = torch.zeros(B, dtype=torch.int32, pin_memory=True)
+        self.h_ulen = torch.zeros(B, dtype=torch.int32, pin_memory=True)
+        self.h_uslot = torch.full((B,), -1, dtype=torch.int32, pin_memory=True)
         # numpy views of the pinned staging (fast per-row fills from lists)
         self.n_ids = self.h_ids.numpy()
         self.n_pos = self.h_pos.numpy()
@@ -67,6 +76,10 @@ class DecodeGraphRunner:
         self.n_bt = self.h_bt.numpy()
         self.n_ctx = self.h_ctx.numpy()
         self.n_lora = self.h_lora.numpy()
+        self.n_order = self.h_order.numpy()
+        self.n_ut0 = self.h_ut0.numpy()
+        self.n_ulen = self.h_ulen.numpy()
+        self.n_uslot = self.h_uslot.numpy()
         # scratch KV block for padding rows = last block of the (extended) pool
         self.scratch_block = runner.config.num_kv_blocks  # extra block id
         self.graphs: Dict[int, torch.cuda.CUDAGraph] = {}
@@ -94,6 +107,10 @@ class DecodeGraphRunner:
             decode_block_tables=self.block_tables[:B],
             decode_context_lens=self.context_lens[:B],
             lora_slot_per_token=self.lora_slots[:B] if self.with_lora else None,
+            lora_order=self.lora_order[:B] if self.with_lora else None,
+            lora_unit_t0=self.lora_ut0[:B] if self.with_lora else None,
+            lora_unit_len=self.lora_ulen[:B] if self.with_lora else None,
+            lora_unit_slot=self.lora_uslot[:B] if self.with_lora else None,
             decode_nsplits=self._graph_nsplits(B),
         )
 
@@ -197,6 +214,20 @@ class DecodeGraphRunner:
         self.block_tables[:B].copy_(self.h_bt[:B], non_blocking=True)
         self.context_lens[:B].copy_(self.h_ctx[:B], non_blocking=True)
         if self.with_lora:
+            from ..lora.manager import build_lora_units
+
+            order, ut0, ulen, uslot = build_lora_units(self.n_lora[:B])
+            u = len(ut0)
+            self.n_order[:B] = order
+            self.n_ut0[:u] = ut0
+            self.n_ulen[:u] = ulen
+            self.n_ulen[u:B] = 0  # padding units: no-op in the kernels
+            self.n_uslot[:u] = uslot
+            self.n_uslot[u:B] = -1
             self.lora_slots[:B].copy_(self.h_lora[:B], non_blocking=True)
+            self.lora_order[:B].copy_(self.h_order[:B], non_blocking=True)
+            self.lora_ut0[:B].copy_(self.h_ut0[:B], non_blocking=True)
+            self.lora_ulen[:B].copy_(self.h_ulen[:B], non_blocking=True)
+            self.lora_uslot[:B].copy_(self.h_uslot[:B], non_blocking=True)
         self.graphs[B].replay()
         return self.logits[B][:n]

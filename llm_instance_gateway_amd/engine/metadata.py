@@ -54,6 +54,13 @@ class BatchMeta:
     lora_tile_t0: Optional[torch.Tensor] = None    # [ntiles] int32
     lora_tile_len: Optional[torch.Tensor] = None   # [ntiles] int32
     lora_tile_slot: Optional[torch.Tensor] = None  # [ntiles] int32
+    # Slot-grouped SGMV units (r02 default path): tokens sorted by adapter
+    # slot (`lora_order`), chunked into <=32-token same-slot units so the
+    # kernels read each adapter's weights once per unit.
+    lora_order: Optional[torch.Tensor] = None      # [T] int32
+    lora_unit_t0: Optional[torch.Tensor] = None    # [U] int32 (into order)
+    lora_unit_len: Optional[torch.Tensor] = None   # [U] int32
+    lora_unit_slot: Optional[torch.Tensor] = None  # [U] int32
 
     @property
     def num_tokens(self) -> int:

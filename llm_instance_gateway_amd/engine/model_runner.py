@@ -262,6 +262,16 @@ class ModelRunner:
             meta.lora_slot_per_token = torch.tensor(slots, dtype=torch.int32).to(
                 dev, non_blocking=True
             )
+            if self.device.type == "cuda":
+                # slot-grouped SGMV units (lora/manager.py::build_lora_units)
+                from ..lora.manager import build_lora_units
+
+                order, ut0, ulen, uslot = build_lora_units(slots)
+                if len(ut0):
+                    meta.lora_order = torch.from_numpy(order).to(dev, non_blocking=True)
+                    meta.lora_unit_t0 = torch.from_numpy(ut0).to(dev, non_blocking=True)
+                    meta.lora_unit_len = torch.from_numpy(ulen).to(dev, non_blocking=True)
+                    meta.lora_unit_slot = torch.from_numpy(uslot).to(dev, non_blocking=True)
             if t0s:
                 meta.lora_tile_t0 = torch.tensor(t0s, dtype=torch.int32).to(
                     dev, non_blocking=True

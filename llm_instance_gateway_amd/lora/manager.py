@@ -39,6 +39,43 @@ log = logging.getLogger(__name__)
 # further work; the wide per-token kernels are the default.
 _TILED_SGMV = os.environ.get("LIG_TILED_SGMV", "0") == "1"
 
+# Slot-grouped unit kernels (r02): default ON; LIG_SGMV_UNITS=0 falls back to
+# the per-token/wide kernels for A/B.
+_UNITS_SGMV = os.environ.get("LIG_SGMV_UNITS", "1") == "1"
+
+def build_lora_units(slots_np, cap: int = 32):
+    """Sort a per-token adapter-slot array into slot-grouped units.
+
+    Returns int32 numpy arrays (order[T], unit_t0[U], unit_len[U],
+    unit_slot[U]): tokens sorted by slot (stable — prefill runs stay
+    contiguous), slot<0 tokens excluded, runs chunked to <= ``cap`` tokens
+    (bounds per-thread serialization in the unit kernels while keeping the
+    adapter-weight read amortized)."""
+    import numpy as np
+
+    arr = np.asarray(slots_np, dtype=np.int32)
+    order = np.argsort(arr, kind="stable").astype(np.int32)
+    sorted_slots = arr[order]
+    t0s, lens, uslots = [], [], []
+    i = int(np.searchsorted(sorted_slots, 0))  # skip slot -1 tokens
+    n = len(arr)
+    while i < n:
+        s = int(sorted_slots[i])
+        k = int(np.searchsorted(sorted_slots, s, side="right"))
+        while i < k:
+            ln = min(cap, k - i)
+            t0s.append(i)
+            lens.append(ln)
+            uslots.append(s)
+            i += ln
+    return (
+        order,
+        np.asarray(t0s, dtype=np.int32),
+        np.asarray(lens, dtype=np.int32),
+        np.asarray(uslots, dtype=np.int32),
+    )
+
+
 # module name -> (in_features, out_features) resolver
 def _module_dims(mc, tp_size: int) -> Dict[str, Tuple[int, int]]:
     h = mc.hidden_size
@@ -103,6 +140,15 @@ class AdapterManager:
                 )
         self._adapters: Dict[str, _Adapter] = {}
         self._slots: List[Optional[str]] = [None] * max_slots
+        # Shrink-output workspace u[T, R]: allocated ONCE at the max flat
+        # batch (token budget rounded up to the GEMM pad multiple).  Decode
+        # hipGraphs bake this pointer, so it must never be reallocated.
+        cfg = runner.config
+        padm = max(getattr(cfg, "pad_batch_multiple", 0), 1)
+        max_t = (cfg.max_num_batched_tokens + padm - 1) // padm * padm
+        self._u_ws = torch.zeros(
+            max(max_t, 256), rank, dtype=self.dtype, device=self.device
+        )
         self._lock = threading.RLock()
         self._side_stream = (
             torch.cuda.Stream(device=self.device)
@@ -332,12 +378,18 @@ class AdapterManager:
         key = (layer_idx, module)
         if ops.native_available() and x.is_cuda:
             T = x.shape[0]
-            ws = getattr(self, "_u_ws", None)
-            if ws is None or ws.shape[0] < T:
-                ws = torch.zeros(
-                    max(T, 4096), self.rank, dtype=self.dtype, device=self.device
+            ws = self._u_ws
+            assert ws.shape[0] >= T, "LoRA workspace smaller than flat batch"
+            # slot-grouped units (r02 default): adapter weights read once per
+            # <=32-token same-slot unit (profiles/r02_kernel_stats.md)
+            order = getattr(meta, "lora_order", None)
+            if order is not None and _UNITS_SGMV:
+                ops._EXT.sgmv_apply_units(
+                    y, x, self.a_stacks[key], self.b_stacks[key], ws,
+                    order, meta.lora_unit_t0, meta.lora_unit_len,
+                    meta.lora_unit_slot, 1.0, ops._stream(),
                 )
-                self._u_ws = ws
+                return
             tile_t0 = getattr(meta, "lora_tile_t0", None)
             if (
                 tile_t0 is not None

@@ -39,6 +39,13 @@ void lig_sgmv_expand_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             hipStream_t);
 void lig_skinny_gemm_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             int, int, int, hipStream_t);
+void lig_sgmv_shrink_units_launch(uint16_t*, const uint16_t*, const uint16_t*,
+                                  const int*, const int*, const int*,
+                                  const int*, int, int, int, hipStream_t);
+void lig_sgmv_expand_units_launch(uint16_t*, const uint16_t*, const uint16_t*,
+                                  const int*, const int*, const int*,
+                                  const int*, float, int, int, int,
+                                  hipStream_t);
 void lig_sgmv_shrink_tiled_launch(uint16_t*, const uint16_t*, const uint16_t*,
                                   const int*, const int*, const int*, int,
                                   int, int, int, hipStream_t);
@@ -245,6 +252,41 @@ void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
                          use_wide, st);
 }
 
+// Slot-grouped units (r02): shrink + expand over host-sorted <=32-token
+// same-slot units through an `order` indirection — adapter weights read once
+// per unit instead of per token (profiles/r02_kernel_stats.md: the per-token
+// kernels were 27% of all kernel time).  Static shapes (padded empty units)
+// so the hipGraph decode path can bake it.
+void sgmv_apply_units(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
+                      torch::Tensor b_stack, torch::Tensor u_ws,
+                      torch::Tensor order, torch::Tensor unit_t0,
+                      torch::Tensor unit_len, torch::Tensor unit_slot,
+                      double scaling, int64_t stream) {
+  TORCH_CHECK(order.scalar_type() == torch::kInt32);
+  TORCH_CHECK(unit_t0.scalar_type() == torch::kInt32);
+  TORCH_CHECK(unit_len.scalar_type() == torch::kInt32);
+  TORCH_CHECK(unit_slot.scalar_type() == torch::kInt32);
+  const int T = x.size(0);
+  const int H = x.size(1);
+  const int R = a_stack.size(1);
+  const int O = y.size(1);
+  const int U = unit_t0.size(0);
+  TORCH_CHECK(u_ws.size(0) >= T && u_ws.size(1) == R, "u workspace too small");
+  TORCH_CHECK(R % 8 == 0 && R <= 64, "rank must be 8/16/32/64");
+  TORCH_CHECK(H % 8 == 0 && O % 8 == 0);
+  if (T == 0 || U == 0) return;
+  hipStream_t st = as_stream(stream);
+  lig_sgmv_shrink_units_launch(
+      bf16_ptr(u_ws), bf16_cptr(x), bf16_cptr(a_stack),
+      order.data_ptr<int>(), unit_t0.data_ptr<int>(),
+      unit_len.data_ptr<int>(), unit_slot.data_ptr<int>(), U, R, H, st);
+  lig_sgmv_expand_units_launch(
+      bf16_ptr(y), bf16_cptr(u_ws), bf16_cptr(b_stack),
+      order.data_ptr<int>(), unit_t0.data_ptr<int>(),
+      unit_len.data_ptr<int>(), unit_slot.data_ptr<int>(), (float)scaling, U,
+      R, O, st);
+}
+
 // Prefill LoRA: one shrink+expand over host-built 64-token segment tiles
 // (MFMA path; engine/model_runner.py builds the tiles).
 void sgmv_apply_tiled(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
@@ -316,6 +358,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_prefill", &paged_attention_prefill);
   m.def("sgmv_shrink", &sgmv_shrink);
   m.def("sgmv_apply", &sgmv_apply);
+  m.def("sgmv_apply_units", &sgmv_apply_units);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("sgmv_apply_tiled", &sgmv_apply_tiled);
   m.def("sgmv_expand", &sgmv_expand);

@@ -364,7 +364,122 @@ __global__ __launch_bounds__(256) void sgmv_expand_tiled_kernel(
   }
 }
 
-// v2 kernels above stay the decode path (mixed per-token slots: more waves
+// ---- slot-grouped "unit" kernels (r02) -------------------------------------
+//
+// r02 steady-state profile (profiles/r02_kernel_stats.md): the four SGMV
+// kernels above are 27% of all kernel time, dominated by re-reading adapter
+// weights per token (expand: B traffic = T*O*R; at T=256 tokens over 32
+// adapters each B is re-read ~8x).  The host sorts the batch's tokens by
+// adapter slot into UNITS of <= 32 same-slot tokens (an indirection array,
+// static-shaped for the hipGraph path) and these kernels read the adapter
+// weights ONCE per unit:
+//
+//   shrink_units: wave = (unit, rank row); 8-token subgroups x 8 H-stripes
+//     like shrink_wide, looping the unit's tokens in chunks of 8 through the
+//     `order` indirection; A row stays L1-hot across the loop.
+//   expand_units: thread = (unit, 8 outputs); B rows loaded once into
+//     registers, then a pipelined loop over the unit's tokens (u load +
+//     y read-modify-write per token; rows are independent).
+
+__global__ void sgmv_shrink_units_kernel(
+    uint16_t* __restrict__ u, const uint16_t* __restrict__ x,
+    const uint16_t* __restrict__ a, const int* __restrict__ order,
+    const int* __restrict__ unit_t0, const int* __restrict__ unit_len,
+    const int* __restrict__ unit_slot, const int U, const int R,
+    const int H) {
+  const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int unit = wid / R;
+  if (unit >= U) return;
+  const int len = unit_len[unit];
+  const int s = unit_slot[unit];
+  if (len <= 0 || s < 0) return;
+  const int r = wid - unit * R;
+  const int t0 = unit_t0[unit];
+  const int lane = threadIdx.x & 63;
+  const int tg = lane >> 3;   // token subgroup 0..7
+  const int li8 = lane & 7;   // H stripe 0..7
+  const int chunks = H / 8;
+  const uint16_t* arow = a + ((size_t)s * R + r) * H;
+
+  for (int j0 = 0; j0 < len; j0 += 8) {
+    const int jj = j0 + tg;
+    const int t = jj < len ? order[t0 + jj] : -1;
+    float acc = 0.0f;
+    if (t >= 0) {
+      const uint16_t* xrow = x + (size_t)t * H;
+#pragma unroll 4
+      for (int c = li8; c < chunks; c += 8) {
+        f32x8 xv =
+            bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(xrow + c * 8));
+        f32x8 av =
+            bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(arow + c * 8));
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc += xv[e] * av[e];
+      }
+    }
+#pragma unroll
+    for (int st = 4; st >= 1; st >>= 1) acc += __shfl_xor(acc, st, 64);
+    if (li8 == 0 && t >= 0) u[(size_t)t * R + r] = f32_to_bf16(acc);
+  }
+}
+
+template <int R>
+__global__ void sgmv_expand_units_kernel(
+    uint16_t* __restrict__ y, const uint16_t* __restrict__ u,
+    const uint16_t* __restrict__ b, const int* __restrict__ order,
+    const int* __restrict__ unit_t0, const int* __restrict__ unit_len,
+    const int* __restrict__ unit_slot, const float scaling, const int U,
+    const int O) {
+  const long total = (long)U * (O / 8);
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += gridDim.x * (long)blockDim.x) {
+    const int oc = idx % (O / 8);
+    const int unit = idx / (O / 8);
+    const int len = unit_len[unit];
+    const int s = unit_slot[unit];
+    if (len <= 0 || s < 0) continue;
+    const int t0 = unit_t0[unit];
+    // B rows for these 8 outputs: loaded ONCE for the whole unit
+    const uint16_t* brow = b + ((size_t)s * O + oc * 8) * R;
+    u16x8 breg[8 * (R / 8)];
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+#pragma unroll
+      for (int rc = 0; rc < R / 8; ++rc)
+        breg[e * (R / 8) + rc] =
+            *reinterpret_cast<const u16x8*>(brow + (size_t)e * R + rc * 8);
+    for (int j = 0; j < len; ++j) {
+      const int t = order[t0 + j];
+      float uv[R];
+#pragma unroll
+      for (int rc = 0; rc < R / 8; ++rc) {
+        f32x8 v = bf16x8_to_f32x8(
+            *reinterpret_cast<const u16x8*>(u + (size_t)t * R + rc * 8));
+#pragma unroll
+        for (int e = 0; e < 8; ++e) uv[rc * 8 + e] = v[e];
+      }
+      u16x8 yv = *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8);
+      f32x8 acc;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] = bf16_to_f32(yv[e]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float aa = 0.0f;
+#pragma unroll
+        for (int rc = 0; rc < R / 8; ++rc) {
+          f32x8 bv = bf16x8_to_f32x8(breg[e * (R / 8) + rc]);
+#pragma unroll
+          for (int q = 0; q < 8; ++q) aa += bv[q] * uv[rc * 8 + q];
+        }
+        acc[e] += aa * scaling;
+      }
+      *reinterpret_cast<u16x8*>(y + (size_t)t * O + oc * 8) =
+          f32x8_to_bf16x8(acc);
+    }
+  }
+}
+
+// v2 kernels above stay the fallback path (no unit metadata: more waves
 // per unit work, no serial reload chains); the _wide variants win only on
 // PREFILL batches, whose tokens are sequence-contiguous so a token group
 // shares one adapter — the caller says which regime this is (`wide`).
@@ -418,6 +533,43 @@ void lig_sgmv_expand_launch(uint16_t* y, const uint16_t* u, const uint16_t* b,
     default: break;  // caller validates R % 8 == 0 && R <= 64
   }
 #undef LAUNCH_E
+}
+
+void lig_sgmv_shrink_units_launch(uint16_t* u, const uint16_t* x,
+                                  const uint16_t* a, const int* order,
+                                  const int* unit_t0, const int* unit_len,
+                                  const int* unit_slot, int U, int R, int H,
+                                  hipStream_t stream) {
+  const long waves = (long)U * R;
+  long grid = (waves + 3) / 4;  // 4 waves per 256-thread block
+  if (grid > 1048576) grid = 1048576;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(sgmv_shrink_units_kernel, dim3((int)grid), dim3(256), 0,
+                     stream, u, x, a, order, unit_t0, unit_len, unit_slot, U,
+                     R, H);
+}
+
+void lig_sgmv_expand_units_launch(uint16_t* y, const uint16_t* u,
+                                  const uint16_t* b, const int* order,
+                                  const int* unit_t0, const int* unit_len,
+                                  const int* unit_slot, float scaling, int U,
+                                  int R, int O, hipStream_t stream) {
+  const long total = (long)U * (O / 8);
+  int grid = (int)((total + 255) / 256);
+  if (grid > 65535) grid = 65535;
+  if (grid < 1) grid = 1;
+#define LAUNCH_EU(RR)                                                       \
+  hipLaunchKernelGGL((sgmv_expand_units_kernel<RR>), dim3(grid), dim3(256), \
+                     0, stream, y, u, b, order, unit_t0, unit_len,          \
+                     unit_slot, scaling, U, O)
+  switch (R) {
+    case 8: LAUNCH_EU(8); break;
+    case 16: LAUNCH_EU(16); break;
+    case 32: LAUNCH_EU(32); break;
+    case 64: LAUNCH_EU(64); break;
+    default: break;
+  }
+#undef LAUNCH_EU
 }
 
 void lig_sgmv_shrink_tiled_launch(uint16_t* u, const uint16_t* x,

@@ -421,3 +421,50 @@ def test_moe_graph_decode_matches_eager():
         del eng
         torch.cuda.empty_cache()
     assert outs[0] == outs[1], f"eager {outs[0]} != graphed {outs[1]}"
+
+
+@pytest.mark.parametrize(
+    "T,H,R,O,pattern",
+    [
+        (256, 4096, 16, 6144, "decode"),   # scattered slots (graph path)
+        (517, 4096, 16, 6144, "prefill"),  # contiguous same-slot runs
+        (96, 64, 8, 128, "decode"),        # tiny/ragged
+        (300, 4096, 32, 4096, "mixed"),    # runs + scattered tail
+    ],
+)
+def test_sgmv_units_matches_ref(T, H, R, O, pattern):
+    """Slot-grouped unit kernels (r02 default SGMV path) vs the per-token
+    fp32 reference."""
+    from llm_instance_gateway_amd.lora.manager import build_lora_units
+
+    torch.manual_seed(31)
+    S = 8
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    a = torch.randn(S, R, H, dtype=torch.bfloat16, device=DEV) * 0.05
+    b = torch.randn(S, O, R, dtype=torch.bfloat16, device=DEV) * 0.05
+    if pattern == "decode":
+        slots = torch.randint(-1, S, (T,), dtype=torch.int32)
+    elif pattern == "prefill":
+        runs, slots_l = [70, 120, 64, 200, 63], []
+        for i, n in enumerate(runs):
+            slots_l += [i % S if i != 2 else -1] * n
+        slots_l += [0] * (T - len(slots_l))
+        slots = torch.tensor(slots_l[:T], dtype=torch.int32)
+    else:
+        slots = torch.cat(
+            [torch.full((150,), 3, dtype=torch.int32),
+             torch.randint(-1, S, (T - 150,), dtype=torch.int32)]
+        )
+    order, ut0, ulen, uslot = build_lora_units(slots.numpy())
+    dev_i = lambda v: torch.as_tensor(v, dtype=torch.int32, device=DEV)
+    y = torch.randn(T, O, dtype=torch.bfloat16, device=DEV) * 0.1
+    y_ref = y.clone().float()
+    ws = torch.zeros(T, R, dtype=torch.bfloat16, device=DEV)
+    ops._EXT.sgmv_apply_units(
+        y, x, a, b, ws, dev_i(order), dev_i(ut0), dev_i(ulen), dev_i(uslot),
+        1.0, torch.cuda.current_stream().cuda_stream,
+    )
+    slots_d = slots.to(DEV)
+    u_ref = torch_ref.sgmv_shrink(x.float(), a.float(), slots_d)
+    torch_ref.sgmv_expand(u_ref, b.float(), slots_d, y_ref, 1.0)
+    close(y, y_ref.to(torch.bfloat16))

@@ -225,3 +225,29 @@ def test_preemption_with_adapter_eviction_interplay():
     roomy = build(num_kv_blocks=128, max_loras=3)
     out_r, _ = run(roomy)
     assert out_c == out_r
+
+
+def test_build_lora_units():
+    """Slot-grouped unit building: stable sort, -1 excluded, runs chunked."""
+    import numpy as np
+
+    from llm_instance_gateway_amd.lora.manager import build_lora_units
+
+    slots = [2, -1, 0, 2, 0, 0, -1, 1]
+    order, t0, ln, us = build_lora_units(slots, cap=2)
+    # sorted slots: [-1,-1,0,0,0,1,2,2]; units over slots>=0 with cap 2
+    assert list(us) == [0, 0, 1, 2]
+    assert list(ln) == [2, 1, 1, 2]
+    assert list(t0) == [2, 4, 5, 6]
+    srt = [slots[i] for i in order]
+    assert srt == sorted(slots)
+    # every unit's tokens share its slot
+    for k in range(len(t0)):
+        toks = order[t0[k]:t0[k] + ln[k]]
+        assert all(slots[t] == us[k] for t in toks)
+    # cap respected at long runs
+    order, t0, ln, us = build_lora_units([3] * 100, cap=32)
+    assert list(ln) == [32, 32, 32, 4]
+    # all -1 -> no units
+    _, t0e, _, _ = build_lora_units([-1, -1])
+    assert len(t0e) == 0
