@@ -5,10 +5,24 @@
 PYTHON ?= python3
 ARCH ?= gfx950
 
-.PHONY: build test test-gpu bench bench-sim bench-router serve clean
+.PHONY: build build-debug test test-gpu test-native-debug bench bench-sim bench-router serve clean
 
 build:
 	PYTORCH_ROCM_ARCH=$(ARCH) $(PYTHON) setup.py build_ext --inplace
+
+# Debug extension build: host-side index/range validation in the bindings
+# (LIG_DEBUG) — catches bad slot/unit/block-table values before a kernel
+# dereferences them.
+build-debug:
+	LIG_DEBUG=1 PYTORCH_ROCM_ARCH=$(ARCH) $(PYTHON) setup.py build_ext --inplace
+
+# Sanitizer pass over the native router core (production logic under
+# ASan+UBSan; SURVEY.md §5 names this gap in the reference).
+test-native-debug:
+	g++ -std=c++17 -O1 -g -fsanitize=address,undefined -fno-omit-frame-pointer \
+	  llm_instance_gateway_amd/router/csrc/router_core_test.cpp \
+	  -o build/router_core_test
+	./build/router_core_test
 
 test:
 	$(PYTHON) -m pytest tests/ -x -q -m "not gpu"

@@ -380,14 +380,18 @@ class AdapterManager:
             T = x.shape[0]
             ws = self._u_ws
             assert ws.shape[0] >= T, "LoRA workspace smaller than flat batch"
-            # slot-grouped units (r02 default): adapter weights read once per
-            # <=32-token same-slot unit (profiles/r02_kernel_stats.md)
+            # slot-grouped units (r02 default): shrink on the occupancy-first
+            # v2 kernels, expand via the LDS-staged unit kernel (B read once
+            # per <=32-token unit; profiles/r02_kernel_stats.md)
             order = getattr(meta, "lora_order", None)
-            if order is not None and _UNITS_SGMV:
+            if order is not None and _UNITS_SGMV and y.shape[1] % 256 == 0:
+                wide = bool(
+                    getattr(meta, "num_prefill_tokens", 0) > 0 and T >= 256
+                )
                 ops._EXT.sgmv_apply_units(
                     y, x, self.a_stacks[key], self.b_stacks[key], ws,
-                    order, meta.lora_unit_t0, meta.lora_unit_len,
-                    meta.lora_unit_slot, 1.0, ops._stream(),
+                    slots, order, meta.lora_unit_t0, meta.lora_unit_len,
+                    meta.lora_unit_slot, 1.0, wide, ops._stream(),
                 )
                 return
             tile_t0 = getattr(meta, "lora_tile_t0", None)

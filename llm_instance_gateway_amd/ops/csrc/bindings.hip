@@ -9,6 +9,25 @@
 
 #include "common.h"
 
+// LIG_DEBUG (make build-debug): host-side validation of device index tensors
+// BEFORE a kernel dereferences them — a D2H copy per check, so debug-only.
+#ifdef LIG_DEBUG
+static void lig_check_range(const torch::Tensor& idx, long lo, long hi,
+                            const char* what) {
+  if (idx.numel() == 0) return;
+  auto c = idx.to(torch::kCPU);
+  const long mn = c.min().item<long>();
+  const long mx = c.max().item<long>();
+  TORCH_CHECK(mn >= lo && mx < hi, "LIG_DEBUG: ", what, " out of range [",
+              lo, ", ", hi, "): min=", mn, " max=", mx);
+}
+#define LIG_CHECK_RANGE(t, lo, hi, what) lig_check_range(t, lo, hi, what)
+#else
+#define LIG_CHECK_RANGE(t, lo, hi, what) \
+  do {                                   \
+  } while (0)
+#endif
+
 extern "C" {
 void lig_rms_norm_launch(uint16_t*, const uint16_t*, const uint16_t*, float,
                          int, int, hipStream_t);
@@ -42,6 +61,10 @@ void lig_skinny_gemm_launch(uint16_t*, const uint16_t*, const uint16_t*,
 void lig_sgmv_shrink_units_launch(uint16_t*, const uint16_t*, const uint16_t*,
                                   const int*, const int*, const int*,
                                   const int*, int, int, int, hipStream_t);
+void lig_sgmv_expand_lds_launch(uint16_t*, const uint16_t*, const uint16_t*,
+                                const int*, const int*, const int*,
+                                const int*, float, int, int, int, int,
+                                hipStream_t);
 void lig_sgmv_expand_units_launch(uint16_t*, const uint16_t*, const uint16_t*,
                                   const int*, const int*, const int*,
                                   const int*, float, int, int, int,
@@ -151,6 +174,9 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(D == 16 || D == 64 || D == 128, "head_dim must be 16/64/128");
   TORCH_CHECK(Hq % Hkv == 0 && Hq / Hkv <= 16, "bad GQA group size");
   if (N == 0) return;
+  LIG_CHECK_RANGE(block_tables, 0, k_cache.size(0), "decode block_tables");
+  LIG_CHECK_RANGE(context_lens, 0, (long)block_tables.size(1) * BS + 1,
+                  "decode context_lens");
   lig_paged_attention_decode_launch(
       bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
       block_tables.data_ptr<int>(), context_lens.data_ptr<int>(),
@@ -244,6 +270,7 @@ void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
   TORCH_CHECK(u_ws.size(0) >= T && u_ws.size(1) == R, "u workspace too small");
   TORCH_CHECK(R % 8 == 0 && R <= 64, "rank must be 8/16/32/64");
   if (T == 0) return;
+  LIG_CHECK_RANGE(slots, -1, a_stack.size(0), "sgmv slots");
   hipStream_t st = as_stream(stream);
   lig_sgmv_shrink_launch(bf16_ptr(u_ws), bf16_cptr(x), bf16_cptr(a_stack),
                          slots.data_ptr<int>(), T, R, H, use_wide, st);
@@ -252,16 +279,20 @@ void sgmv_apply(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
                          use_wide, st);
 }
 
-// Slot-grouped units (r02): shrink + expand over host-sorted <=32-token
-// same-slot units through an `order` indirection — adapter weights read once
-// per unit instead of per token (profiles/r02_kernel_stats.md: the per-token
-// kernels were 27% of all kernel time).  Static shapes (padded empty units)
-// so the hipGraph decode path can bake it.
+// Slot-grouped units (r02): shrink via the proven v2 kernels (occupancy-
+// first), expand via the LDS-staged unit kernel — full per-token thread
+// parallelism with B read once per <=32-token unit instead of per token
+// (profiles/r02_kernel_stats.md: the per-token kernels were 27% of all
+// kernel time).  Static shapes (padded empty units) so the hipGraph decode
+// path can bake it.  Requires O % 256 == 0 (every serving module dim);
+// caller falls back to sgmv_apply otherwise.
 void sgmv_apply_units(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
                       torch::Tensor b_stack, torch::Tensor u_ws,
-                      torch::Tensor order, torch::Tensor unit_t0,
-                      torch::Tensor unit_len, torch::Tensor unit_slot,
-                      double scaling, int64_t stream) {
+                      torch::Tensor slots, torch::Tensor order,
+                      torch::Tensor unit_t0, torch::Tensor unit_len,
+                      torch::Tensor unit_slot, double scaling, bool use_wide,
+                      int64_t stream) {
+  TORCH_CHECK(slots.scalar_type() == torch::kInt32);
   TORCH_CHECK(order.scalar_type() == torch::kInt32);
   TORCH_CHECK(unit_t0.scalar_type() == torch::kInt32);
   TORCH_CHECK(unit_len.scalar_type() == torch::kInt32);
@@ -273,18 +304,21 @@ void sgmv_apply_units(torch::Tensor y, torch::Tensor x, torch::Tensor a_stack,
   const int U = unit_t0.size(0);
   TORCH_CHECK(u_ws.size(0) >= T && u_ws.size(1) == R, "u workspace too small");
   TORCH_CHECK(R % 8 == 0 && R <= 64, "rank must be 8/16/32/64");
-  TORCH_CHECK(H % 8 == 0 && O % 8 == 0);
+  TORCH_CHECK(H % 8 == 0 && O % 256 == 0, "units expand needs O%256==0");
   if (T == 0 || U == 0) return;
+  LIG_CHECK_RANGE(slots, -1, a_stack.size(0), "sgmv slots");
+  LIG_CHECK_RANGE(order, 0, T, "sgmv unit order");
+  LIG_CHECK_RANGE(unit_t0, 0, T, "sgmv unit_t0");
+  LIG_CHECK_RANGE(unit_len, 0, T + 1, "sgmv unit_len");
+  LIG_CHECK_RANGE(unit_slot, -1, b_stack.size(0), "sgmv unit_slot");
   hipStream_t st = as_stream(stream);
-  lig_sgmv_shrink_units_launch(
-      bf16_ptr(u_ws), bf16_cptr(x), bf16_cptr(a_stack),
-      order.data_ptr<int>(), unit_t0.data_ptr<int>(),
-      unit_len.data_ptr<int>(), unit_slot.data_ptr<int>(), U, R, H, st);
-  lig_sgmv_expand_units_launch(
+  lig_sgmv_shrink_launch(bf16_ptr(u_ws), bf16_cptr(x), bf16_cptr(a_stack),
+                         slots.data_ptr<int>(), T, R, H, use_wide, st);
+  lig_sgmv_expand_lds_launch(
       bf16_ptr(y), bf16_cptr(u_ws), bf16_cptr(b_stack),
       order.data_ptr<int>(), unit_t0.data_ptr<int>(),
       unit_len.data_ptr<int>(), unit_slot.data_ptr<int>(), (float)scaling, U,
-      R, O, st);
+      T, R, O, st);
 }
 
 // Prefill LoRA: one shrink+expand over host-built 64-token segment tiles

@@ -371,15 +371,83 @@ __global__ __launch_bounds__(256) void sgmv_expand_tiled_kernel(
 // weights per token (expand: B traffic = T*O*R; at T=256 tokens over 32
 // adapters each B is re-read ~8x).  The host sorts the batch's tokens by
 // adapter slot into UNITS of <= 32 same-slot tokens (an indirection array,
-// static-shaped for the hipGraph path) and these kernels read the adapter
-// weights ONCE per unit:
+// static-shaped for the hipGraph path).
 //
-//   shrink_units: wave = (unit, rank row); 8-token subgroups x 8 H-stripes
-//     like shrink_wide, looping the unit's tokens in chunks of 8 through the
-//     `order` indirection; A row stays L1-hot across the loop.
-//   expand_units: thread = (unit, 8 outputs); B rows loaded once into
-//     registers, then a pipelined loop over the unit's tokens (u load +
-//     y read-modify-write per token; rows are independent).
+// v1 units kernels (register-resident B, thread = (unit, 8 outs) with a
+// serial token loop) were a MEASURED DEAD END: 31.0 vs 52.6 req/s e2e
+// (gpurun bench_units vs bench_nounits) — they traded occupancy for
+// locality (decode: 384-512 waves on a 256-CU chip) and the adapter set is
+// L2-resident anyway, so the binding constraint was LATENCY, exactly what
+// the v2 kernels' design note says.  Kept below for the record.
+//
+// expand_lds (the shipping units path) keeps FULL per-token parallelism and
+// amortizes B through LDS: workgroup = (unit, 256-output chunk); the B tile
+// (256 rows x R, a CONTIGUOUS 256*R-element span of the row-major stack) is
+// cooperatively staged in LDS once, then 8-token x 32-output-chunk threads
+// sweep the unit's tokens in passes of 8, each doing one y RMW against the
+// LDS tile.  B traffic drops T/cap-fold with the same thread count as the
+// per-token kernel.
+
+template <int R>
+__global__ __launch_bounds__(256) void sgmv_expand_lds_kernel(
+    uint16_t* __restrict__ y, const uint16_t* __restrict__ u,
+    const uint16_t* __restrict__ b, const int* __restrict__ order,
+    const int* __restrict__ unit_t0, const int* __restrict__ unit_len,
+    const int* __restrict__ unit_slot, const float scaling, const int T,
+    const int O) {
+  const int unit = blockIdx.y;
+  const int len = unit_len[unit];
+  const int s = unit_slot[unit];
+  if (len <= 0 || s < 0) return;
+  const int t0 = unit_t0[unit];
+  const int o0 = blockIdx.x * 256;
+  const int tid = threadIdx.x;
+
+  // stage B rows [o0, o0+256) x R — contiguous in the row-major stack
+  __shared__ uint16_t bs[256 * R];
+  const uint16_t* bsrc = b + ((size_t)s * O + o0) * R;
+  constexpr int VECS = 256 * R / 8;  // u16x8 vectors
+#pragma unroll
+  for (int v = tid; v < VECS; v += 256)
+    *reinterpret_cast<u16x8*>(bs + v * 8) =
+        *reinterpret_cast<const u16x8*>(bsrc + (size_t)v * 8);
+  __syncthreads();
+
+  const int jt = tid >> 5;   // token lane 0..7 within a pass
+  const int oc = tid & 31;   // 8-output chunk 0..31 within the 256 block
+  for (int j0 = 0; j0 < len; j0 += 8) {
+    const int j = j0 + jt;
+    if (j >= len) break;
+    const int t = order[t0 + j];
+    float uv[R];
+#pragma unroll
+    for (int rc = 0; rc < R / 8; ++rc) {
+      f32x8 v = bf16x8_to_f32x8(
+          *reinterpret_cast<const u16x8*>(u + (size_t)t * R + rc * 8));
+#pragma unroll
+      for (int e = 0; e < 8; ++e) uv[rc * 8 + e] = v[e];
+    }
+    uint16_t* yp = y + (size_t)t * O + o0 + oc * 8;
+    u16x8 yv = *reinterpret_cast<u16x8*>(yp);
+    f32x8 acc;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = bf16_to_f32(yv[e]);
+    const uint16_t* brow = bs + (size_t)(oc * 8) * R;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float aa = 0.0f;
+#pragma unroll
+      for (int rc = 0; rc < R / 8; ++rc) {
+        f32x8 bv = bf16x8_to_f32x8(
+            *reinterpret_cast<const u16x8*>(brow + (size_t)e * R + rc * 8));
+#pragma unroll
+        for (int q = 0; q < 8; ++q) aa += bv[q] * uv[rc * 8 + q];
+      }
+      acc[e] += aa * scaling;
+    }
+    *reinterpret_cast<u16x8*>(yp) = f32x8_to_bf16x8(acc);
+  }
+}
 
 __global__ void sgmv_shrink_units_kernel(
     uint16_t* __restrict__ u, const uint16_t* __restrict__ x,
@@ -533,6 +601,26 @@ void lig_sgmv_expand_launch(uint16_t* y, const uint16_t* u, const uint16_t* b,
     default: break;  // caller validates R % 8 == 0 && R <= 64
   }
 #undef LAUNCH_E
+}
+
+void lig_sgmv_expand_lds_launch(uint16_t* y, const uint16_t* u,
+                                const uint16_t* b, const int* order,
+                                const int* unit_t0, const int* unit_len,
+                                const int* unit_slot, float scaling, int U,
+                                int T, int R, int O, hipStream_t stream) {
+  dim3 grid(O / 256, U);
+#define LAUNCH_EL(RR)                                                        \
+  hipLaunchKernelGGL((sgmv_expand_lds_kernel<RR>), grid, dim3(256), 0,       \
+                     stream, y, u, b, order, unit_t0, unit_len, unit_slot,   \
+                     scaling, T, O)
+  switch (R) {
+    case 8: LAUNCH_EL(8); break;
+    case 16: LAUNCH_EL(16); break;
+    case 32: LAUNCH_EL(32); break;
+    case 64: LAUNCH_EL(64); break;
+    default: break;
+  }
+#undef LAUNCH_EL
 }
 
 void lig_sgmv_shrink_units_launch(uint16_t* u, const uint16_t* x,

@@ -28,16 +28,21 @@ sources = sorted(
     if not p.name.endswith("_hip.hip")
 )
 
+# LIG_DEBUG=1: host-side index/range validation in the bindings (make
+# build-debug) — slot ids vs stack sizes, unit tables vs batch bounds —
+# catching bad inputs before a kernel dereferences them.
+debug_flags = ["-DLIG_DEBUG=1", "-g"] if os.environ.get("LIG_DEBUG") == "1" else []
+
 ext = CUDAExtension(
     name="llm_instance_gateway_amd._C",
     sources=sources,
     extra_compile_args={
-        "cxx": ["-O3", "-std=c++17"],
+        "cxx": ["-O3", "-std=c++17"] + debug_flags,
         "nvcc": [
             "-O3",
             "-std=c++17",
             "--offload-arch=gfx950",
-        ],
+        ] + debug_flags,
     },
 )
 

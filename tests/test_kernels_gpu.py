@@ -428,7 +428,7 @@ def test_moe_graph_decode_matches_eager():
     [
         (256, 4096, 16, 6144, "decode"),   # scattered slots (graph path)
         (517, 4096, 16, 6144, "prefill"),  # contiguous same-slot runs
-        (96, 64, 8, 128, "decode"),        # tiny/ragged
+        (96, 64, 8, 256, "decode"),        # small/ragged
         (300, 4096, 32, 4096, "mixed"),    # runs + scattered tail
     ],
 )
@@ -460,11 +460,12 @@ def test_sgmv_units_matches_ref(T, H, R, O, pattern):
     y = torch.randn(T, O, dtype=torch.bfloat16, device=DEV) * 0.1
     y_ref = y.clone().float()
     ws = torch.zeros(T, R, dtype=torch.bfloat16, device=DEV)
-    ops._EXT.sgmv_apply_units(
-        y, x, a, b, ws, dev_i(order), dev_i(ut0), dev_i(ulen), dev_i(uslot),
-        1.0, torch.cuda.current_stream().cuda_stream,
-    )
     slots_d = slots.to(DEV)
+    ops._EXT.sgmv_apply_units(
+        y, x, a, b, ws, slots_d, dev_i(order), dev_i(ut0), dev_i(ulen),
+        dev_i(uslot), 1.0, pattern == "prefill" and T >= 256,
+        torch.cuda.current_stream().cuda_stream,
+    )
     u_ref = torch_ref.sgmv_shrink(x.float(), a.float(), slots_d)
     torch_ref.sgmv_expand(u_ref, b.float(), slots_d, y_ref, 1.0)
     close(y, y_ref.to(torch.bfloat16))
