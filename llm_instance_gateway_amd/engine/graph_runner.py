@@ -58,6 +58,9 @@ class DecodeGraphRunner:
         self.lora_ut0 = torch.zeros(B, dtype=torch.int32, device=dev)
         self.lora_ulen = torch.zeros(B, dtype=torch.int32, device=dev)
         self.lora_uslot = torch.full((B,), -1, dtype=torch.int32, device=dev)
+        # graph-private LoRA shrink workspace (see metadata.lora_ws): replay
+        # must not share u with a concurrently-running eager prefill
+        self.lora_ws: Optional[torch.Tensor] = None
         # pinned host staging (one copy per tensor per step)
         self.h_ids = torch.zeros(B, dtype=torch.int64, pin_memory=True)
         self.h_pos = torch.zeros(B, dtype=torch.int64, pin_memory=True)
@@ -111,6 +114,7 @@ class DecodeGraphRunner:
             lora_unit_t0=self.lora_ut0[:B] if self.with_lora else None,
             lora_unit_len=self.lora_ulen[:B] if self.with_lora else None,
             lora_unit_slot=self.lora_uslot[:B] if self.with_lora else None,
+            lora_ws=self.lora_ws,
             decode_nsplits=self._graph_nsplits(B),
         )
 
@@ -122,6 +126,11 @@ class DecodeGraphRunner:
 
     def capture(self, with_lora: bool) -> None:
         self.with_lora = with_lora
+        if with_lora and self.lora_ws is None:
+            mgr = self.runner.adapter_manager
+            self.lora_ws = torch.zeros(
+                self.max_bs, mgr.rank, dtype=mgr.dtype, device=self.device
+            )
         # Preallocate the flash-decode split scratch at the max any bucket
         # needs so every captured graph bakes one stable pointer (the ops
         # cache grows monotonically; growth during capture would leave

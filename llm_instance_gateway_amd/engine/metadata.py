@@ -61,6 +61,9 @@ class BatchMeta:
     lora_unit_t0: Optional[torch.Tensor] = None    # [U] int32 (into order)
     lora_unit_len: Optional[torch.Tensor] = None   # [U] int32
     lora_unit_slot: Optional[torch.Tensor] = None  # [U] int32
+    # Private LoRA shrink workspace for this batch (the decode graph owns its
+    # own so graph replay and a concurrent eager prefill never share u).
+    lora_ws: Optional[torch.Tensor] = None
 
     @property
     def num_tokens(self) -> int:

@@ -103,12 +103,18 @@ class ModelRunner:
         # MoE decode is graph-safe too: decode-only batches use the dense
         # expert dispatch (static shapes, no host sync — models/mixtral.py).
         self._use_graphs = self.device.type == "cuda" and not config.enforce_eager
+        # Mixed-step prefill/decode stream overlap (LIG_OVERLAP=0 disables).
+        import os as _os
+
+        self._overlap = _os.environ.get("LIG_OVERLAP", "1") == "1"
+        self._decode_stream = None
 
     # -- batch construction ---------------------------------------------------
-    def _build_batch(self, out: SchedulerOutput):
+    def _build_batch(self, out):
+        scheduled = out.scheduled if isinstance(out, SchedulerOutput) else out
         prefills: List[ScheduledSeq] = []
         decodes: List[ScheduledSeq] = []
-        for ss in out.scheduled:
+        for ss in scheduled:
             (decodes if ss.num_tokens == 1 else prefills).append(ss)
 
         bs = self.config.block_size
@@ -299,6 +305,10 @@ class ModelRunner:
                 res = self._execute_graph(out)
             if res is not None:
                 return res
+        if self._use_graphs and self._overlap:
+            res = self._execute_overlapped(out)
+            if res is not None:
+                return res
         with gpu_range("build_batch"):
             ids, pos, meta, sample_seqs, sample_idx, ordered = self._build_batch(out)
         with gpu_range("forward"):
@@ -315,19 +325,72 @@ class ModelRunner:
             ss.seq.num_computed_tokens += ss.num_tokens
         return sampled
 
-    def _execute_graph(self, out: SchedulerOutput) -> Optional[Dict[int, int]]:
-        """Decode-only fast path: one hipGraph replay for the whole step."""
+    def _execute_overlapped(self, out: SchedulerOutput) -> Optional[Dict[int, int]]:
+        """Mixed-step stream overlap (ROADMAP r01 #9): the decode partition
+        replays its hipGraph on a side stream while the prefill chunk runs
+        eager on the main stream.  Safe by construction: KV slots are
+        disjoint, adapter slot weights are read-only during a step, and the
+        graph owns a private LoRA workspace (metadata.lora_ws)."""
+        prefills = [ss for ss in out.scheduled if ss.num_tokens > 1]
+        decodes = [ss for ss in out.scheduled if ss.num_tokens == 1]
+        if not prefills or not decodes:
+            return None
+        gr = self._ensure_graph_runner()
+        if gr is None or gr.pick_size(len(decodes)) is None:
+            return None
+        if self._decode_stream is None:
+            self._decode_stream = torch.cuda.Stream(device=self.device)
+        sd = self._decode_stream
+        main = torch.cuda.current_stream(self.device)
+        slot_of = (
+            self.adapter_manager.slot_of if self.adapter_manager is not None else None
+        )
+        with gpu_range("overlap_step"):
+            sd.wait_stream(main)  # adapter load fences etc. ride main
+            with torch.cuda.stream(sd):
+                logits_d = gr.run(decodes, slot_of)
+            ids, pos, meta, sample_seqs_p, sample_idx_p, _ = self._build_batch(
+                prefills
+            )
+            hidden = self.model(ids, pos, self.kv_caches, meta)
+            logits_p = None
+            if sample_seqs_p:
+                idx = torch.tensor(
+                    sample_idx_p, dtype=torch.int64, device=self.device
+                )
+                logits_p = self.model.compute_logits(hidden, idx)
+            main.wait_stream(sd)
+            sampled: Dict[int, int] = {}
+            if logits_p is not None:
+                for seq, tok in zip(
+                    sample_seqs_p, sample_tokens(logits_p, sample_seqs_p)
+                ):
+                    sampled[seq.seq_id] = tok
+            d_seqs = [ss.seq for ss in decodes]
+            for seq, tok in zip(d_seqs, sample_tokens(logits_d.float(), d_seqs)):
+                sampled[seq.seq_id] = tok
+        for ss in out.scheduled:
+            ss.seq.num_computed_tokens += ss.num_tokens
+        return sampled
+
+    def _ensure_graph_runner(self):
         from .graph_runner import DecodeGraphRunner
 
-        decodes = out.scheduled
         want_lora = self.adapter_manager is not None
         if self.graph_runner is None or self.graph_runner.with_lora != want_lora:
             self.graph_runner = DecodeGraphRunner(self)
             self.graph_runner.capture(with_lora=want_lora)
-        gr = self.graph_runner
+        return self.graph_runner
+
+    def _execute_graph(self, out: SchedulerOutput) -> Optional[Dict[int, int]]:
+        """Decode-only fast path: one hipGraph replay for the whole step."""
+        decodes = out.scheduled
+        gr = self._ensure_graph_runner()
         if gr.pick_size(len(decodes)) is None:
             return None  # larger than any captured bucket: eager fallback
-        slot_of = self.adapter_manager.slot_of if want_lora else None
+        slot_of = (
+            self.adapter_manager.slot_of if self.adapter_manager is not None else None
+        )
         logits = gr.run(decodes, slot_of)
         seqs = [ss.seq for ss in decodes]
         tokens = sample_tokens(logits.float(), seqs)

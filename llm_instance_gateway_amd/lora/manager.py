@@ -378,7 +378,9 @@ class AdapterManager:
         key = (layer_idx, module)
         if ops.native_available() and x.is_cuda:
             T = x.shape[0]
-            ws = self._u_ws
+            ws = getattr(meta, "lora_ws", None)
+            if ws is None:
+                ws = self._u_ws
             assert ws.shape[0] >= T, "LoRA workspace smaller than flat batch"
             # slot-grouped units (r02 default): shrink on the occupancy-first
             # v2 kernels, expand via the LDS-staged unit kernel (B read once
