@@ -31,8 +31,13 @@
 // un-normalized partial (oacc, m, l) goes to scratch fp32
 // [N, Hq, nsplits, D+2] and a combine kernel merges the slices.  Doubles
 // wave-level parallelism for long contexts (ROADMAP #1).
-template <int HEAD_DIM, bool SPLIT = false>
-__global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
+// MAX_THREADS: real blocks are GROUP*64 threads (GQA group <= 16 -> <=1024),
+// but Llama/Mixtral run GROUP=4 (256 threads); compiling the hot
+// instantiation with launch_bounds(256) lifts the 1024-thread VGPR ceiling
+// (110 VGPRs -> 4 waves/SIMD at LB=1024) so the scheduler can trade
+// registers for occupancy per the real block size.
+template <int HEAD_DIM, bool SPLIT = false, int MAX_THREADS = 1024>
+__global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
     uint16_t* __restrict__ out,            // [N, Hq, D]
     const uint16_t* __restrict__ q,        // [N, Hq, D]
     const uint16_t* __restrict__ k_cache,  // [B, Hkv, BS, D]
@@ -92,20 +97,22 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
     it1 = min(iters, it0 + span);
   }
 
-  // Software-pipelined K/V: the next iteration's vector loads are issued
+  // Software-pipelined K/V: upcoming iterations' vector loads are issued
   // BEFORE this iteration's dot/softmax/accumulate chain (whose xor-shuffle
   // reduces serialize ~10 cross-lane ops), so HBM latency hides behind
-  // compute.  The two prefetch buffers are SEPARATE NAMED arrays selected at
-  // compile time by a manually 2-unrolled loop: a runtime-indexed buf[2][NV]
-  // spills to scratch (first attempt measured 6x slower than no prefetch).
+  // compute.  THREE prefetch buffers as SEPARATE NAMED arrays rotated by a
+  // manually 3-unrolled loop: a runtime-indexed buf[][NV] spills to scratch
+  // (measured 6x slower), and the third stage costs +16 VGPRs (110 -> 126),
+  // still within the 4-waves/SIMD occupancy bin — deeper latency hiding for
+  // free.
   constexpr int NV = DL >= 8 ? DL / 8 : 1;  // u16x8 vectors per lane
-  u16x8 kraw0[NV], vraw0[NV], kraw1[NV], vraw1[NV];
-  bool valid0 = false, valid1 = false;
+  u16x8 kraw0[NV], vraw0[NV], kraw1[NV], vraw1[NV], kraw2[NV], vraw2[NV];
+  bool valid0 = false, valid1 = false, valid2 = false;
 
 #define ISSUE_LOAD(KBUF, VBUF, VALID, IT)                                    \
   do {                                                                       \
     const int t_ = (IT) * 8 + tok;                                           \
-    (VALID) = t_ < ctx;                                                      \
+    (VALID) = (IT) < it1 && t_ < ctx;                                        \
     if (VALID) {                                                             \
       const int blk_ = bt[t_ / block_size];                                  \
       const size_t base_ = (((size_t)blk_ * num_kv_heads + kvh) *            \
@@ -172,13 +179,17 @@ __global__ __launch_bounds__(1024) void paged_attention_decode_kernel(
         oacc[i] = oacc[i] * alpha + p * vf[i];                               \
   } while (0)
 
-  if (it1 > it0) ISSUE_LOAD(kraw0, vraw0, valid0, it0);
-  for (int it = it0; it < it1; it += 2) {
-    if (it + 1 < it1) ISSUE_LOAD(kraw1, vraw1, valid1, it + 1);
+  ISSUE_LOAD(kraw0, vraw0, valid0, it0);
+  ISSUE_LOAD(kraw1, vraw1, valid1, it0 + 1);
+  for (int it = it0; it < it1; it += 3) {
+    ISSUE_LOAD(kraw2, vraw2, valid2, it + 2);
     STEP_BODY(kraw0, vraw0, valid0);
     if (it + 1 >= it1) break;
-    if (it + 2 < it1) ISSUE_LOAD(kraw0, vraw0, valid0, it + 2);
+    ISSUE_LOAD(kraw0, vraw0, valid0, it + 3);
     STEP_BODY(kraw1, vraw1, valid1);
+    if (it + 2 >= it1) break;
+    ISSUE_LOAD(kraw1, vraw1, valid1, it + 4);
+    STEP_BODY(kraw2, vraw2, valid2);
   }
 #undef ISSUE_LOAD
 #undef STEP_BODY
@@ -275,11 +286,18 @@ extern "C" void lig_paged_attention_decode_split_launch(
   dim3 block(group * WAVE_SIZE);
 #define LAUNCH_S(HD)                                                        \
   do {                                                                      \
-    hipLaunchKernelGGL((paged_attention_decode_kernel<HD, true>), grid,     \
-                       block, 0, stream, out, q, k_cache, v_cache,          \
-                       block_tables, context_lens, scale, num_q_heads,      \
-                       num_kv_heads, block_size, max_blocks, scratch,       \
-                       nsplits);                                            \
+    if (group <= 4)                                                         \
+      hipLaunchKernelGGL((paged_attention_decode_kernel<HD, true, 256>),    \
+                         grid, block, 0, stream, out, q, k_cache, v_cache,  \
+                         block_tables, context_lens, scale, num_q_heads,    \
+                         num_kv_heads, block_size, max_blocks, scratch,     \
+                         nsplits);                                          \
+    else                                                                    \
+      hipLaunchKernelGGL((paged_attention_decode_kernel<HD, true>), grid,   \
+                         block, 0, stream, out, q, k_cache, v_cache,        \
+                         block_tables, context_lens, scale, num_q_heads,    \
+                         num_kv_heads, block_size, max_blocks, scratch,     \
+                         nsplits);                                          \
     hipLaunchKernelGGL((paged_attention_decode_combine_kernel<HD>),         \
                        dim3(num_seqs, num_q_heads), dim3(WAVE_SIZE), 0,     \
                        stream, out, scratch, num_q_heads, nsplits);         \
@@ -303,10 +321,18 @@ extern "C" void lig_paged_attention_decode_launch(
   dim3 grid(num_seqs, num_kv_heads);
   dim3 block(group * WAVE_SIZE);
 #define LAUNCH(HD)                                                          \
-  hipLaunchKernelGGL((paged_attention_decode_kernel<HD>), grid, block, 0,   \
-                     stream, out, q, k_cache, v_cache, block_tables,        \
-                     context_lens, scale, num_q_heads, num_kv_heads,        \
-                     block_size, max_blocks)
+  do {                                                                      \
+    if (group <= 4)                                                         \
+      hipLaunchKernelGGL((paged_attention_decode_kernel<HD, false, 256>),   \
+                         grid, block, 0, stream, out, q, k_cache, v_cache,  \
+                         block_tables, context_lens, scale, num_q_heads,    \
+                         num_kv_heads, block_size, max_blocks);             \
+    else                                                                    \
+      hipLaunchKernelGGL((paged_attention_decode_kernel<HD>), grid, block,  \
+                         0, stream, out, q, k_cache, v_cache, block_tables, \
+                         context_lens, scale, num_q_heads, num_kv_heads,    \
+                         block_size, max_blocks);                           \
+  } while (0)
   switch (head_dim) {
     case 16: LAUNCH(16); break;
     case 64: LAUNCH(64); break;
