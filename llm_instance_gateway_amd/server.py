@@ -88,6 +88,11 @@ def main(argv: Optional[List[str]] = None) -> int:
         help="route with the C++ filter tree reading shared-memory gauges "
         "(default: Python tree; both are semantically identical)",
     )
+    ap.add_argument(
+        "--tokenizer", default="",
+        help="HuggingFace tokenizer.json for the front door (default: the "
+        "asset-free byte codec)",
+    )
     ap.add_argument("--v", type=int, default=1, help="log verbosity (klog analogue)")
     args = ap.parse_args(argv)
 
@@ -98,6 +103,13 @@ def main(argv: Optional[List[str]] = None) -> int:
 
     from .api.config import ConfigReconciler, load_config
     from .serving.frontdoor import build_app
+    from .utils import tokenizer as _tok
+
+    if args.tokenizer:
+        if _tok.load_backend(args.tokenizer):
+            log.info("front-door tokenizer: %s", _tok.backend_name())
+        else:
+            log.warning("could not load tokenizer %s; using byte codec", args.tokenizer)
 
     cfg = load_config(args.config) if args.config else None
 

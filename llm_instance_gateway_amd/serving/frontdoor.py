@@ -350,7 +350,7 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
 
     @app.post("/v1/chat/completions")
     async def chat_completions(req: ChatCompletionRequest):
-        prompt = "\n".join(f"{m.role}: {m.content}" for m in req.messages)
+        prompt = tokenizer.apply_chat_template(req.messages)
         creq = CompletionRequest(
             model=req.model,
             prompt=prompt,

@@ -39,3 +39,41 @@ def test_gpu_range_noop_on_cpu():
             pass  # must not raise without a GPU
     finally:
         set_gpu_tracing(False)
+
+
+def test_tokenizer_real_backend_roundtrip(tmp_path):
+    """A real HF vocabulary plugs into the front-door codec (the vLLM API
+    surface the reference pins); built in-test since no assets ship offline."""
+    from tokenizers import Tokenizer, models, pre_tokenizers, trainers
+
+    from llm_instance_gateway_amd.utils import tokenizer as tok
+
+    t = Tokenizer(models.BPE(unk_token="<unk>"))
+    t.pre_tokenizer = pre_tokenizers.Whitespace()
+    trainer = trainers.BpeTrainer(special_tokens=["<unk>"], vocab_size=200)
+    t.train_from_iterator(
+        ["hello world", "hello gateway", "routing requests to engines"] * 8,
+        trainer,
+    )
+    path = tmp_path / "tokenizer.json"
+    t.save(str(path))
+    try:
+        assert tok.load_backend(str(path))
+        assert tok.backend_name() == "Tokenizer"
+        ids = tok.encode("hello world")
+        assert ids and ids != [tok.BOS_ID] + [b + tok.OFFSET for b in b"hello world"]
+        assert "hello" in tok.decode(ids)
+    finally:
+        tok.set_backend(None)  # restore byte codec for other tests
+    # byte codec restored
+    assert tok.decode(tok.encode("abc", add_bos=False)) == "abc"
+
+
+def test_chat_template_shape():
+    from llm_instance_gateway_amd.utils import tokenizer as tok
+
+    out = tok.apply_chat_template(
+        [{"role": "user", "content": "hi"}], add_generation_prompt=True
+    )
+    assert "<|start_header_id|>user<|end_header_id|>" in out
+    assert out.endswith("assistant<|end_header_id|>\n\n")
