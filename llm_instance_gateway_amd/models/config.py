@@ -31,6 +31,10 @@ class ModelConfig:
     # MoE (Mixtral-style); num_experts == 0 means dense.
     num_experts: int = 0
     num_experts_per_tok: int = 2
+    # Multi-GPU MoE mode: False = TP-shard every expert's intermediate dim;
+    # True = expert parallelism (each rank owns num_experts/world full
+    # experts; models/mixtral.py MoEBlock docstring).
+    expert_parallel: bool = False
     dtype: str = "bfloat16"
 
     @property

@@ -6,11 +6,12 @@ SwiGLU MLP is replaced by a top-k routed expert block.
 
 MI355X design notes:
 
-  * Experts are **TP-sharded** (every rank holds a 1/tp slice of every
-    expert's intermediate dim) rather than expert-parallel: with 288 GB HBM
-    per GPU the full 8x7B expert set fits on one device, so EP's all-to-all
-    over xGMI is not needed for parity, and the per-block all-reduce pattern
-    stays identical to the dense model.
+  * Two multi-GPU modes: **TP-sharded** experts (default: every rank holds a
+    1/tp slice of every expert's intermediate dim) and **expert-parallel**
+    (``ModelConfig.expert_parallel``: each rank owns E/world full experts —
+    see MoEBlock docstring for why the combine stays one all-reduce over
+    xGMI rather than an all-to-all in this replicated-activation design).
+    With 288 GB HBM the full 8x7B expert set also fits one device (TP=1).
   * Dispatch is sort-by-expert into a capacity-padded [E, cap, h] batch and
     ONE bmm pair (w13 fused gate+up, then w2) through hipBLASLt — the
     per-expert Python loop was host-launch-bound (E x 3 tiny ops x layers).
@@ -44,16 +45,40 @@ from .llama import Attention, _linear
 
 
 class MoEBlock(nn.Module):
-    """Top-k routed experts with fused gate+up per expert."""
+    """Top-k routed experts with fused gate+up per expert.
+
+    Parallel modes over the engine's process group:
+      * TP (default): every rank holds a 1/tp slice of EVERY expert's
+        intermediate dim; combine = the block all-reduce.
+      * EP (``cfg.expert_parallel``): each rank OWNS num_experts/world full
+        experts and computes partial token outputs for the experts it owns;
+        the same all-reduce combines the partials.  This is the xGMI-shaped
+        expert parallelism for this engine model: activations are replicated
+        across the TP group by construction (every rank runs the identical
+        batch), so the classic all-to-all token exchange would move data that
+        is already resident — the collective stays ONE all-reduce per block
+        while each expert's GEMMs run at full intermediate width.  A2A-based
+        dispatch only pays when activations are sharded across ranks (a
+        DP-coupled fleet), which this design avoids.
+    """
 
     def __init__(self, cfg: ModelConfig, device, dtype, tp=None) -> None:
         super().__init__()
         self.tp = tp
         tp_size = tp.world_size if tp is not None else 1
-        assert cfg.intermediate_size % tp_size == 0
+        self.ep = bool(getattr(cfg, "expert_parallel", False)) and tp_size > 1
+        if self.ep:
+            assert cfg.num_experts % tp_size == 0, "EP needs E % world == 0"
+            self.local_experts = cfg.num_experts // tp_size
+            self.expert_base = tp.rank * self.local_experts
+            inter = cfg.intermediate_size  # full width per owned expert
+        else:
+            assert cfg.intermediate_size % tp_size == 0
+            self.local_experts = cfg.num_experts
+            self.expert_base = 0
+            inter = cfg.intermediate_size // tp_size
         self.num_experts = cfg.num_experts
         self.top_k = cfg.num_experts_per_tok
-        inter = cfg.intermediate_size // tp_size
         h = cfg.hidden_size
         # router is replicated (tiny GEMM, identical on every rank)
         self.gate = _linear(h, cfg.num_experts, device, dtype)
@@ -63,11 +88,11 @@ class MoEBlock(nn.Module):
         # [8,1073,4096]x[8,4096,28672]^T on MI355X; contiguous operands are
         # also the faster layout).
         self.w13 = nn.Parameter(
-            torch.empty(cfg.num_experts, h, 2 * inter, device=device, dtype=dtype)
+            torch.empty(self.local_experts, h, 2 * inter, device=device, dtype=dtype)
             .uniform_(-0.02, 0.02)
         )
         self.w2 = nn.Parameter(
-            torch.empty(cfg.num_experts, inter, h, device=device, dtype=dtype)
+            torch.empty(self.local_experts, inter, h, device=device, dtype=dtype)
             .uniform_(-0.02, 0.02)
         )
 
@@ -89,7 +114,10 @@ class MoEBlock(nn.Module):
                 T, self.num_experts, dtype=x.dtype, device=x.device
             )
             wdense.scatter_(1, experts, weights)
-            E = self.num_experts
+            # EP: this rank computes only its owned expert slice; the block
+            # all-reduce sums the partials
+            wdense = wdense[:, self.expert_base:self.expert_base + self.local_experts]
+            E = self.local_experts
             xe = x.unsqueeze(0).expand(E, T, h)
             act = ops.silu_and_mul(torch.bmm(xe, self.w13).view(E * T, -1))
             ye = torch.bmm(act.view(E, T, -1), self.w2)  # [E, T, h]
@@ -107,12 +135,21 @@ class MoEBlock(nn.Module):
             torch.arange(T, device=x.device).unsqueeze(1).expand(T, self.top_k).reshape(-1)
         )
         flat_w = weights.reshape(-1)
+        if self.ep:
+            # keep only (token, expert) pairs owned by this rank; partial
+            # outputs are summed by the block all-reduce
+            mine = (flat_experts >= self.expert_base) & (
+                flat_experts < self.expert_base + self.local_experts
+            )
+            flat_experts = flat_experts[mine] - self.expert_base
+            flat_tokens = flat_tokens[mine]
+            flat_w = flat_w[mine]
         # sort the (token, expert) pairs by expert -> contiguous per-expert runs
         order = torch.argsort(flat_experts, stable=True)
         sorted_experts = flat_experts[order]
         sorted_tokens = flat_tokens[order]
         sorted_w = flat_w[order]
-        counts = torch.bincount(sorted_experts, minlength=self.num_experts)
+        counts = torch.bincount(sorted_experts, minlength=self.local_experts)
 
         # Capacity-padded batched dispatch: ONE bmm pair for all experts
         # instead of E small GEMMs + Python loop (the loop was host-bound:
@@ -121,11 +158,13 @@ class MoEBlock(nn.Module):
         # eager, so there is no graph to break).  Padding rows index a zero
         # row appended to x and scatter back with weight 0 — bit-exact vs
         # the per-expert loop.
-        cap = int(counts.max())
+        cap = int(counts.max()) if counts.numel() else 0
         if cap == 0:
-            return x.new_zeros(T, h)
+            out = x.new_zeros(T, h)
+            # EP ranks with no owned tokens still join the combine
+            return self.tp.all_reduce(out) if self.tp is not None else out
         starts = torch.cumsum(counts, 0) - counts
-        E = self.num_experts
+        E = self.local_experts
         dev = x.device
         # position of each sorted entry within its expert's run
         pos_in_e = (
