@@ -1,9 +1,17 @@
 #!/usr/bin/env python3
-"""One-off GEMM tuning run (GPU): sweeps hipBLASLt algorithms for every GEMM
-shape the serving step hits, writing llm_instance_gateway_amd/data/
-tunableop_gfx950.csv (committed; loaded read-only by every engine at startup).
+"""One-off GEMM tuning run (GPU): sweeps hipBLASLt algorithms for the EAGER
+serving GEMM shapes, writing llm_instance_gateway_amd/data/
+tunableop_gfx950.csv (committed; loaded by every engine at startup and
+enabled ONLY around eager forwards — graph capture/replay keeps the default
+algorithms, which avoids the r01-measured capture regression).
 
-    python benchmarks/tune_gemms.py [--model llama3-8b] [--steps 6]
+Eager GEMMs are the mixed prefill+decode steps, whose flat batch is bucketed
+to multiples of 256 tokens (engine/config.py pad_batch_multiple), so the
+shape set is closed: M in {256..4096 step 256} x the four projection GEMMs.
+Decode-only steps run inside hipGraphs where TunableOp stays disabled, so
+their shapes need no tuning.
+
+    python benchmarks/tune_gemms.py [--model llama3-8b]
 """
 
 import argparse
@@ -17,47 +25,53 @@ sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="llama3-8b")
-    ap.add_argument("--steps", type=int, default=30)
-    ap.add_argument("--concurrency", type=int, default=128)
+    ap.add_argument("--max-m", type=int, default=4096)
+    ap.add_argument("--step-m", type=int, default=256)
     args = ap.parse_args()
 
     os.environ["LIG_TUNE_GEMMS"] = "1"
     import torch
 
-    from llm_instance_gateway_amd.utils.tunableop import RESULTS_FILE, maybe_enable_tunableop
+    from llm_instance_gateway_amd.models.config import get_model_config
+    from llm_instance_gateway_amd.utils.tunableop import (
+        RESULTS_FILE,
+        maybe_enable_tunableop,
+    )
 
     assert torch.cuda.is_available()
     assert maybe_enable_tunableop()
 
-    from llm_instance_gateway_amd.engine import EngineConfig, LLMEngine, SamplingParams
-    from llm_instance_gateway_amd.models.config import get_model_config
-
-    cfg = EngineConfig(
-        model=get_model_config(args.model),
-        num_kv_blocks=8192,
-        max_num_seqs=256,
-        max_num_batched_tokens=4096,
-        max_model_len=2048,
-        max_loras=0,
-        enforce_eager=True,  # capture would freeze pre-tuning algorithms
-    )
+    mc = get_model_config(args.model)
+    h = mc.hidden_size
+    q = mc.num_heads * mc.head_dim
+    kv = mc.num_kv_heads * mc.head_dim
+    shapes = [
+        (h, q + 2 * kv),            # qkv_proj
+        (q, h),                     # o_proj
+    ]
+    if not mc.is_moe:
+        shapes += [
+            (h, 2 * mc.intermediate_size),  # gate_up
+            (mc.intermediate_size, h),      # down
+        ]
+    dev = "cuda:0"
     torch.manual_seed(0)
-    eng = LLMEngine(cfg, device="cuda:0")
-    g = torch.Generator().manual_seed(1)
-    # short prompts: a couple of prefill steps, then a long run of M=concurrency
-    # decode steps — the decode GEMM shapes (M = graph bucket = concurrency)
-    # are the stable, highest-leverage shapes to tune
-    for i in range(args.concurrency):
-        plen = int(torch.randint(8, 24, (1,), generator=g))
-        eng.add_request(
-            torch.randint(1, 128000, (plen,), generator=g).tolist(),
-            SamplingParams(max_new_tokens=args.steps + 8),
-        )
-    for _ in range(args.steps):
-        if not eng.has_work():
-            break
-        eng.step()
-    torch.cuda.synchronize()
+    weights = {
+        (k_in, n_out): torch.randn(n_out, k_in, dtype=torch.bfloat16, device=dev)
+        * 0.01
+        for (k_in, n_out) in shapes
+    }
+    ms = list(range(args.step_m, args.max_m + 1, args.step_m))
+    total = len(ms) * len(shapes)
+    done = 0
+    for m in ms:
+        for (k_in, n_out) in shapes:
+            x = torch.randn(m, k_in, dtype=torch.bfloat16, device=dev)
+            for _ in range(3):
+                torch.nn.functional.linear(x, weights[(k_in, n_out)])
+            torch.cuda.synchronize()
+            done += 1
+            print(f"[{done}/{total}] tuned M={m} K={k_in} N={n_out}", flush=True)
     print(f"tuning pass done; results flush to {RESULTS_FILE} at exit")
 
 

@@ -15,6 +15,7 @@ from typing import Dict, List, Optional, Tuple
 import torch
 
 from ..models.config import ModelConfig
+from ..utils.tunableop import eager_tuned
 from .config import EngineConfig
 from .metadata import BatchMeta
 from ..utils.tracing import gpu_range
@@ -311,7 +312,7 @@ class ModelRunner:
                 return res
         with gpu_range("build_batch"):
             ids, pos, meta, sample_seqs, sample_idx, ordered = self._build_batch(out)
-        with gpu_range("forward"):
+        with gpu_range("forward"), eager_tuned():
             hidden = self.model(ids, pos, self.kv_caches, meta)
         sampled: Dict[int, int] = {}
         if sample_seqs:
@@ -352,7 +353,8 @@ class ModelRunner:
             ids, pos, meta, sample_seqs_p, sample_idx_p, _ = self._build_batch(
                 prefills
             )
-            hidden = self.model(ids, pos, self.kv_caches, meta)
+            with eager_tuned():
+                hidden = self.model(ids, pos, self.kv_caches, meta)
             logits_p = None
             if sample_seqs_p:
                 idx = torch.tensor(
