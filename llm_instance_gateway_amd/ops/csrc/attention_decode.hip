@@ -97,17 +97,18 @@ __global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
     it1 = min(iters, it0 + span);
   }
 
-  // Software-pipelined K/V: upcoming iterations' vector loads are issued
+  // Software-pipelined K/V: the next iteration's vector loads are issued
   // BEFORE this iteration's dot/softmax/accumulate chain (whose xor-shuffle
   // reduces serialize ~10 cross-lane ops), so HBM latency hides behind
-  // compute.  THREE prefetch buffers as SEPARATE NAMED arrays rotated by a
-  // manually 3-unrolled loop: a runtime-indexed buf[][NV] spills to scratch
-  // (measured 6x slower), and the third stage costs +16 VGPRs (110 -> 126),
-  // still within the 4-waves/SIMD occupancy bin — deeper latency hiding for
-  // free.
+  // compute.  TWO prefetch buffers as SEPARATE NAMED arrays selected by a
+  // manually 2-unrolled loop: a runtime-indexed buf[2][NV] spills to scratch
+  // (measured 6x slower than no prefetch).  A THIRD stage was a MEASURED
+  // DEAD END: +34 VGPRs (110 -> 144) dropped occupancy 4 -> 3 waves/SIMD
+  // and decode bandwidth fell 2.6 -> 1.9 TB/s (gpurun bench_attention A/B);
+  // occupancy beats pipeline depth here.
   constexpr int NV = DL >= 8 ? DL / 8 : 1;  // u16x8 vectors per lane
-  u16x8 kraw0[NV], vraw0[NV], kraw1[NV], vraw1[NV], kraw2[NV], vraw2[NV];
-  bool valid0 = false, valid1 = false, valid2 = false;
+  u16x8 kraw0[NV], vraw0[NV], kraw1[NV], vraw1[NV];
+  bool valid0 = false, valid1 = false;
 
 #define ISSUE_LOAD(KBUF, VBUF, VALID, IT)                                    \
   do {                                                                       \
@@ -180,16 +181,12 @@ __global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
   } while (0)
 
   ISSUE_LOAD(kraw0, vraw0, valid0, it0);
-  ISSUE_LOAD(kraw1, vraw1, valid1, it0 + 1);
-  for (int it = it0; it < it1; it += 3) {
-    ISSUE_LOAD(kraw2, vraw2, valid2, it + 2);
+  for (int it = it0; it < it1; it += 2) {
+    ISSUE_LOAD(kraw1, vraw1, valid1, it + 1);
     STEP_BODY(kraw0, vraw0, valid0);
     if (it + 1 >= it1) break;
-    ISSUE_LOAD(kraw0, vraw0, valid0, it + 3);
+    ISSUE_LOAD(kraw0, vraw0, valid0, it + 2);
     STEP_BODY(kraw1, vraw1, valid1);
-    if (it + 2 >= it1) break;
-    ISSUE_LOAD(kraw1, vraw1, valid1, it + 4);
-    STEP_BODY(kraw2, vraw2, valid2);
   }
 #undef ISSUE_LOAD
 #undef STEP_BODY
