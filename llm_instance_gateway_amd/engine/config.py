@@ -29,6 +29,11 @@ class EngineConfig:
     # Preempt (recompute) newest sequences above this KV utilisation
     # (reference continous_batching.py:117-131).
     preempt_kv_threshold: float = 0.90
+    # Mixed steps (default): decodes run EVERY step, prefill chunks fill the
+    # remaining token budget (vLLM chunked-prefill semantics; lets the runner
+    # overlap the graphed decode bucket with the eager prefill).  False keeps
+    # the reference sim's prefill_or_decode alternation.
+    mixed_steps: bool = True
     # Multi-LoRA slots (vLLM --max-loras analogue).
     max_loras: int = 8
     max_lora_rank: int = 16

@@ -146,15 +146,32 @@ class EngineScheduler:
 
     # -- the step builder -------------------------------------------------------
     def schedule(self) -> SchedulerOutput:
-        """One step is EITHER all-prefill-chunks OR all-decodes (the
-        reference simulation's prefill_or_decode loop,
-        continous_batching.py:173-191, and what lets decode steps replay a
-        captured hipGraph — see engine/graph_runner.py).  Prefill work is
-        prioritised (TTFT); decodes run when no prefill is admittable."""
-        out = self._schedule_prefills()
-        if not out.is_empty:
-            return out
-        return self._schedule_decodes(out)
+        """Build one step.
+
+        Default (``mixed_steps``): decodes are scheduled EVERY step (one
+        token per running sequence) and prefill chunks fill the remaining
+        token budget — vLLM's chunked-prefill semantics.  Decode-only steps
+        still replay hipGraphs; mixed steps run the decode bucket's graph on
+        a side stream overlapped with the eager prefill
+        (engine/model_runner.py::_execute_overlapped), so a prefill chunk no
+        longer stalls every in-flight decode for a full forward (~15 ms TPOT
+        spikes at serving shapes).
+
+        ``mixed_steps=False`` keeps the reference simulation's
+        prefill_or_decode alternation (continous_batching.py:173-191):
+        all-prefill steps prioritised, decodes only when no prefill is
+        admittable."""
+        if not self.config.mixed_steps:
+            out = self._schedule_prefills()
+            if not out.is_empty:
+                return out
+            return self._schedule_decodes(out)
+        out = SchedulerOutput()
+        self._schedule_decodes(out)
+        budget = self.config.max_num_batched_tokens - len(out.scheduled)
+        if budget > 0:
+            self._schedule_prefills(out, budget)
+        return out
 
     def _schedule_decodes(self, out: SchedulerOutput) -> SchedulerOutput:
         cfg = self.config
@@ -190,11 +207,17 @@ class EngineScheduler:
                 continue
         return out
 
-    def _schedule_prefills(self) -> SchedulerOutput:
-        out = SchedulerOutput()
+    def _schedule_prefills(
+        self, out: Optional[SchedulerOutput] = None, budget: Optional[int] = None
+    ) -> SchedulerOutput:
+        if out is None:
+            out = SchedulerOutput()
         cfg = self.config
         bm = self.block_manager
-        token_budget = cfg.max_num_batched_tokens
+        token_budget = budget if budget is not None else cfg.max_num_batched_tokens
+        # sequences preempted earlier in THIS step must not be re-admitted in
+        # the same step (their blocks were freed to make room)
+        preempted_now = {s.seq_id for s in out.preempted}
 
         # 1) Continue chunked prefills already running.
         for seq in list(self.running):
@@ -225,6 +248,9 @@ class EngineScheduler:
             and bm.usage < cfg.preempt_kv_threshold
         ):
             seq = self.waiting[skip]
+            if seq.seq_id in preempted_now:
+                skip += 1
+                continue
             if (
                 seq.lora_name
                 and cfg.max_loras > 0
