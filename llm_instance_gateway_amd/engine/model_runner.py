@@ -104,10 +104,16 @@ class ModelRunner:
         # MoE decode is graph-safe too: decode-only batches use the dense
         # expert dispatch (static shapes, no host sync — models/mixtral.py).
         self._use_graphs = self.device.type == "cuda" and not config.enforce_eager
-        # Mixed-step prefill/decode stream overlap (LIG_OVERLAP=0 disables).
+        # Mixed-step prefill/decode stream overlap — MEASURED DEAD END as a
+        # default: 51.4 vs 56.5 req/s at the serving shape (gpurun
+        # bench_mixed vs bench_mixed_noov) — the graph replay and the eager
+        # prefill contend for the same CUs, so splitting the step into two
+        # streams only added sync overhead; the single flat mixed batch
+        # already keeps the chip full.  Kept opt-in (LIG_OVERLAP=1) for
+        # shapes where decode is small vs prefill.
         import os as _os
 
-        self._overlap = _os.environ.get("LIG_OVERLAP", "1") == "1"
+        self._overlap = _os.environ.get("LIG_OVERLAP", "0") == "1"
         self._decode_stream = None
 
     # -- batch construction ---------------------------------------------------
