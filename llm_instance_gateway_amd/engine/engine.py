@@ -76,6 +76,13 @@ class LLMEngine:
                 f"prompt of {len(prompt_token_ids)} tokens exceeds max_model_len "
                 f"{self.config.max_model_len}"
             )
+        # vLLM semantics: generation never runs past the model's context
+        # window — clamp max_new_tokens to the remaining budget (otherwise a
+        # long request walks positions past the RoPE cache / KV capacity).
+        if sampling is not None:
+            room = self.config.max_model_len - len(prompt_token_ids)
+            if sampling.max_new_tokens > room:
+                sampling.max_new_tokens = max(1, room)
         seq = Sequence(
             prompt_token_ids,
             sampling=sampling,

@@ -283,3 +283,29 @@ def test_mixed_steps_combine_prefill_and_decode():
         for ss in out.scheduled:
             tok = 1  # appended via engine.step normally; here just advance
         del eng
+
+
+def test_max_new_tokens_clamped_to_context_window():
+    """Generation never walks positions past max_model_len (vLLM clamp
+    semantics) — unclamped it would run off the RoPE cache / KV capacity."""
+    import torch
+
+    from llm_instance_gateway_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from llm_instance_gateway_amd.models.config import get_model_config
+
+    torch.manual_seed(0)
+    eng = LLMEngine(
+        EngineConfig(
+            model=get_model_config("tiny"),
+            num_kv_blocks=64,
+            max_num_seqs=4,
+            max_num_batched_tokens=128,
+            max_model_len=64,
+            max_loras=0,
+        ),
+        device="cpu",
+    )
+    seq = eng.add_request(list(range(1, 51)), SamplingParams(max_new_tokens=500))
+    assert seq.sampling.max_new_tokens == 14  # 64 - 50
+    eng.generate_all()
+    assert seq.total_len <= 64
