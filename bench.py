@@ -141,6 +141,7 @@ def main() -> None:
     stats = {"done": 0, "good": 0, "good_lo": 0, "good_hi": 0, "tokens": 0}
     ttfts: list = []
     tpots: list = []
+    queues: list = []  # arrival -> first scheduled (admission wait)
 
     primed = [False]
 
@@ -173,6 +174,8 @@ def main() -> None:
                         ttfts.append(ttft)
                     if tpot is not None:
                         tpots.append(tpot)
+                    if seq.first_scheduled_time is not None:
+                        queues.append(seq.first_scheduled_time - seq.arrival_time)
                     if ttft is not None and ttft <= TTFT_SLO_S:
                         stats["good"] += 1
                         if tpot is not None and tpot <= TPOT_SLO_LO_S:
@@ -281,6 +284,8 @@ def main() -> None:
                 "tpot_p50_s": pct(tpots, 0.50),
                 "tpot_p95_s": pct(tpots, 0.95),
                 "tpot_p99_s": pct(tpots, 0.99),
+                "queue_p50_s": pct(queues, 0.50),
+                "queue_p95_s": pct(queues, 0.95),
             },
         }
         print(json.dumps(result))
