@@ -142,6 +142,7 @@ def main() -> None:
     ttfts: list = []
     tpots: list = []
     queues: list = []  # arrival -> first scheduled (admission wait)
+    trace: list = []   # LIG_BENCH_TRACE=1: per-request tail diagnosis
 
     primed = [False]
 
@@ -175,7 +176,14 @@ def main() -> None:
                     if tpot is not None:
                         tpots.append(tpot)
                     if seq.first_scheduled_time is not None:
-                        queues.append(seq.first_scheduled_time - seq.arrival_time)
+                        q = seq.first_scheduled_time - seq.arrival_time
+                        queues.append(q)
+                        if os.environ.get("LIG_BENCH_TRACE") == "1":
+                            trace.append(
+                                (q, seq.arrival_time, seq.first_scheduled_time,
+                                 seq.first_token_time, seq.prompt_len,
+                                 seq.num_output_tokens, seq.num_preemptions)
+                            )
                     if ttft is not None and ttft <= TTFT_SLO_S:
                         stats["good"] += 1
                         if tpot is not None and tpot <= TPOT_SLO_LO_S:
@@ -289,6 +297,15 @@ def main() -> None:
             },
         }
         print(json.dumps(result))
+        if trace:
+            t_ref = min(t[1] for t in trace)
+            trace.sort(reverse=True)
+            print("# worst queue waits (q_s, arrive_rel_s, sched_rel_s, "
+                  "ttft_rel_s, plen, out, preempts):", file=sys.stderr)
+            for q, a, sc, ft, pl, no, pre in trace[:12]:
+                print(f"#  {q:8.3f} {a - t_ref:8.3f} {sc - t_ref:8.3f} "
+                      f"{(ft - t_ref) if ft else -1:8.3f} {pl:4d} {no:4d} {pre}",
+                      file=sys.stderr)
     if distributed:
         torch.distributed.destroy_process_group()
 
