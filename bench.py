@@ -145,17 +145,20 @@ def main() -> None:
     trace: list = []   # LIG_BENCH_TRACE=1: per-request tail diagnosis
 
     primed = [False]
+    flood: list = []  # the initial simultaneous fill (cold-start population)
 
     def top_up():
         in_flight = eng.scheduler.num_waiting + eng.scheduler.num_running
         prime = not primed[0]
         while in_flight < args.concurrency:
             prompt, olen, lora = gen(prime=prime)
-            eng.add_request(
+            seq = eng.add_request(
                 prompt,
                 SamplingParams(max_new_tokens=olen),
                 lora_name=lora if args.lora else "",
             )
+            if prime:
+                flood.append(seq)
             in_flight += 1
         primed[0] = True
 
@@ -195,23 +198,25 @@ def main() -> None:
     # -- warmup (untimed) ----------------------------------------------------
     # Contract floor: W untimed steps.  Then warm BY STATE: the closed loop
     # starts with `concurrency` simultaneous arrivals whose prefill flood
-    # dominates any short timed window (r01's 5.95 vs 32.9 req/s defect), so
-    # keep stepping until that initial population has fully turned over
-    # (completions >= concurrency).  After turnover every in-flight request
-    # arrived via steady-state replacement, so a timed window of ANY length
-    # samples the steady state.
+    # dominates any short timed window (r01's 5.95 vs 32.9 req/s defect).
+    # Keep stepping until EVERY member of that initial flood has finished —
+    # any survivor would carry its cold-start admission wait (measured: the
+    # last flood prompts admit ~2 s after t=0 at conc=256) into the timed
+    # window's TTFT tail.  After that, every in-flight request arrived via
+    # steady-state replacement, so a timed window of ANY length samples the
+    # steady state.
     run_steps(args.warmup, record=False)
-    warm_done = 0
     warm_extra = 0
     warm_cap_steps = 4096
     warm_deadline = time.monotonic() + 150.0
     while (
-        warm_done < args.concurrency
+        any(not s.is_finished() for s in flood)
         and warm_extra < warm_cap_steps
         and time.monotonic() < warm_deadline
     ):
-        warm_done += run_steps(8, record=False)
+        run_steps(8, record=False)
         warm_extra += 8
+    flood.clear()  # release the references
 
     if distributed:
         torch.distributed.barrier()
