@@ -28,7 +28,7 @@ from .metadata import BatchMeta
 
 log = logging.getLogger(__name__)
 
-CAPTURE_SIZES = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256]
+CAPTURE_SIZES = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256, 320, 384, 448, 512]
 
 
 class DecodeGraphRunner:
