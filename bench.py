@@ -74,7 +74,10 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=128)
     ap.add_argument("--warmup", type=int, default=32)
     ap.add_argument("--model", default="llama3-8b")
-    ap.add_argument("--concurrency", type=int, default=256)
+    # conc sweep on MI355X (gpurun_out/sw_c*.log): 256→54.2, 384→67.0,
+    # 512→67.0 req/s with worsening TPOT; 384 saturates the chip
+    # (~25.4k tok/s) with TTFT p95 ~44 ms and is the default operating point.
+    ap.add_argument("--concurrency", type=int, default=384)
     ap.add_argument("--device", default=None, help="override (e.g. cpu for dev)")
     ap.add_argument("--max-batched-tokens", type=int, default=0,
                     help="prefill chunk budget override (0 = default)")
