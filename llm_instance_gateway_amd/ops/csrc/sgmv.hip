@@ -403,14 +403,24 @@ __global__ __launch_bounds__(256) void sgmv_expand_lds_kernel(
   const int o0 = blockIdx.x * 256;
   const int tid = threadIdx.x;
 
-  // stage B rows [o0, o0+256) x R — contiguous in the row-major stack
-  __shared__ uint16_t bs[256 * R];
-  const uint16_t* bsrc = b + ((size_t)s * O + o0) * R;
-  constexpr int VECS = 256 * R / 8;  // u16x8 vectors
+  // Stage B rows [o0, o0+256) x R TRANSPOSED: bs[r][o].  The row-major
+  // [o][r] layout put thread oc's inner-loop reads at byte offset oc*2R*8 —
+  // a multiple of 256 B for R=16, i.e. every lane of a half-wave on the
+  // SAME LDS bank (32-way conflict on every B read; the kernel sat at
+  // ~1 TB/s, 5x over its traffic floor).  Transposed, lane oc reads the
+  // 16 B span bs[r*256 + oc*8] — consecutive lanes, consecutive banks,
+  // conflict-free.  Staging: thread tid transposes output row o0+tid with
+  // b16 writes of stride 256 elements (2-way write conflicts, once per WG).
+  __shared__ uint16_t bs[R * 256];
+  {
+    const uint16_t* brow = b + ((size_t)s * O + o0 + tid) * R;
 #pragma unroll
-  for (int v = tid; v < VECS; v += 256)
-    *reinterpret_cast<u16x8*>(bs + v * 8) =
-        *reinterpret_cast<const u16x8*>(bsrc + (size_t)v * 8);
+    for (int rc = 0; rc < R / 8; ++rc) {
+      u16x8 v = *reinterpret_cast<const u16x8*>(brow + rc * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) bs[(rc * 8 + e) * 256 + tid] = v[e];
+    }
+  }
   __syncthreads();
 
   const int jt = tid >> 5;   // token lane 0..7 within a pass
@@ -431,21 +441,19 @@ __global__ __launch_bounds__(256) void sgmv_expand_lds_kernel(
     u16x8 yv = *reinterpret_cast<u16x8*>(yp);
     f32x8 acc;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) acc[e] = bf16_to_f32(yv[e]);
-    const uint16_t* brow = bs + (size_t)(oc * 8) * R;
+    for (int e = 0; e < 8; ++e) acc[e] = 0.0f;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      float aa = 0.0f;
+    for (int r = 0; r < R; ++r) {
+      f32x8 bv = bf16x8_to_f32x8(
+          *reinterpret_cast<const u16x8*>(bs + (size_t)r * 256 + oc * 8));
+      const float us = uv[r];
 #pragma unroll
-      for (int rc = 0; rc < R / 8; ++rc) {
-        f32x8 bv = bf16x8_to_f32x8(
-            *reinterpret_cast<const u16x8*>(brow + (size_t)e * R + rc * 8));
-#pragma unroll
-        for (int q = 0; q < 8; ++q) aa += bv[q] * uv[rc * 8 + q];
-      }
-      acc[e] += aa * scaling;
+      for (int e = 0; e < 8; ++e) acc[e] += bv[e] * us;
     }
-    *reinterpret_cast<u16x8*>(yp) = f32x8_to_bf16x8(acc);
+    f32x8 out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) out[e] = bf16_to_f32(yv[e]) + acc[e] * scaling;
+    *reinterpret_cast<u16x8*>(yp) = f32x8_to_bf16x8(out);
   }
 }
 
