@@ -153,8 +153,13 @@ __global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
     _Pragma("unroll") for (int i = 0; i < DL; ++i) s += qreg[i] * kf[i];     \
     s = group_reduce_sum<8>(s);                                              \
     s = valid ? s * scale : -INFINITY;                                       \
-    const float chunk_max = stride_reduce_max<8, 8>(s);                      \
-    const float new_m = fmaxf(m, chunk_max);                                 \
+    /* PER-TOKEN-STREAM online softmax: the 8 lanes of this token group    */\
+    /* keep their own (m, l, oacc) for the tokens they stream (tok, tok+8, */\
+    /* ...).  The old per-iteration cross-token reduces (stride_reduce_max */\
+    /* + stride_reduce_sum = 6 serialized shuffles EVERY 8-token chunk)    */\
+    /* were the kernel's latency chain; the 8 states merge ONCE after the  */\
+    /* K loop instead.                                                     */\
+    const float new_m = fmaxf(m, s);                                         \
     float p, alpha;                                                          \
     if (new_m == -INFINITY) {                                                \
       p = 0.0f;                                                              \
@@ -164,7 +169,7 @@ __global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
       p = (s == -INFINITY) ? 0.0f : __expf(s - new_m);                       \
       m = new_m;                                                             \
     }                                                                        \
-    l = l * alpha + stride_reduce_sum<8, 8>(p);                              \
+    l = l * alpha + p;                                                       \
     float vf[DL];                                                            \
     if constexpr (DL >= 8) {                                                 \
       _Pragma("unroll") for (int i = 0; i < NV; ++i) {                       \
@@ -191,9 +196,16 @@ __global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
 #undef ISSUE_LOAD
 #undef STEP_BODY
 
-  // combine the 8 token-subsets (lanes stride-8 apart share dchunk)
+  // Merge the 8 per-token-stream softmax states (lanes stride-8 apart share
+  // dchunk): rescale every stream to the global max, then sum l and oacc.
+  {
+    const float gm = stride_reduce_max<8, 8>(m);
+    const float w = (m == -INFINITY) ? 0.0f : __expf(m - gm);
+    l = stride_reduce_sum<8, 8>(l * w);
 #pragma unroll
-  for (int i = 0; i < DL; ++i) oacc[i] = stride_reduce_sum<8, 8>(oacc[i]);
+    for (int i = 0; i < DL; ++i) oacc[i] = stride_reduce_sum<8, 8>(oacc[i] * w);
+    m = gm;
+  }
 
   if constexpr (SPLIT) {
     // un-normalized partial out; m and l are wave-uniform after the
