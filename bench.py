@@ -79,6 +79,13 @@ def main() -> None:
     # (~25.4k tok/s) with TTFT p95 ~44 ms and is the default operating point.
     ap.add_argument("--concurrency", type=int, default=384)
     ap.add_argument("--device", default=None, help="override (e.g. cpu for dev)")
+    ap.add_argument(
+        "--quant",
+        default="",
+        choices=["", "fp8"],
+        help="opt-in fp8 (e4m3) projection GEMMs; the headline metric stays "
+        "bf16 (BASELINE's dtype) — fp8 runs are reported with dtype=fp8",
+    )
     ap.add_argument("--max-batched-tokens", type=int, default=0,
                     help="prefill chunk budget override (0 = default)")
     ap.add_argument(
@@ -127,6 +134,7 @@ def main() -> None:
         max_loras=NUM_LORAS,
         engine_id=f"engine-{rank}",
         gpu_id=local_rank,
+        quant=args.quant,
     )
     torch.manual_seed(1234)  # identical weights across ranks
     eng = LLMEngine(cfg, device=device)
@@ -282,7 +290,7 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": "bf16",
+            "dtype": "fp8" if args.quant == "fp8" else "bf16",
             "data": "synthetic (random-init weights, Poisson-shaped 202±20 in / 179±17 out)",
             "config": {
                 "model": args.model,

@@ -34,6 +34,10 @@ class EngineConfig:
     # overlap the graphed decode bucket with the eager prefill).  False keeps
     # the reference sim's prefill_or_decode alternation.
     mixed_steps: bool = True
+    # Opt-in quantization: "" (bf16, default) or "fp8" (e4m3 weights +
+    # dynamic per-token activations on the projection GEMMs — models/quant.py;
+    # vLLM --quantization fp8 analogue).  Dense models only.
+    quant: str = ""
     # Multi-LoRA slots (vLLM --max-loras analogue).
     max_loras: int = 8
     max_lora_rank: int = 16

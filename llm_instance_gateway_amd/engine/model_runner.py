@@ -75,6 +75,15 @@ class ModelRunner:
         mc = config.model
         log.info("building model %s on %s", mc.name, self.device)
         self.model = build_model(mc, self.device, dtype, tp)
+        if config.quant == "fp8":
+            if mc.is_moe:
+                raise ValueError("quant='fp8' covers dense models only")
+            from ..models.quant import convert_linears_to_fp8
+
+            n_q = convert_linears_to_fp8(self.model)
+            if self.device.type == "cuda":
+                torch.cuda.empty_cache()  # bf16 projection weights are gone
+            log.info("fp8 quantization: %d projection GEMMs converted", n_q)
         tp_size = tp.world_size if tp is not None else 1
         if self.device.type == "cuda":
             n = determine_num_kv_blocks(config, self.device, tp_size)

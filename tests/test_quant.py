@@ -1,0 +1,126 @@
+"""fp8 (e4m3) quantization semantics — CPU.
+
+The CPU path of models/quant.py runs the SAME quantized values through an
+fp32 matmul, so these tests pin the quantization scheme itself; the GPU
+_scaled_mm path (tests/test_quant_gpu.py) must then agree with it up to
+accumulation order.
+"""
+
+import pytest
+import torch
+
+from llm_instance_gateway_amd.models.quant import (
+    FP8_MAX,
+    Fp8Linear,
+    convert_linears_to_fp8,
+    quantize_fp8_rowwise,
+)
+
+
+def test_quantize_roundtrip_error_bounded():
+    torch.manual_seed(0)
+    w = torch.randn(64, 256, dtype=torch.bfloat16)
+    q, s = quantize_fp8_rowwise(w)
+    assert q.dtype == torch.float8_e4m3fn and s.shape == (64, 1)
+    deq = q.float() * s
+    # e4m3 has a 3-bit mantissa: elementwise relative error <= 2^-4 at the
+    # top binade; allow 2x slack for values far below the row amax
+    err = (deq - w.float()).abs()
+    rel = err / w.float().abs().clamp(min=1e-3)
+    assert rel.median() < 0.05
+    assert (deq.abs() <= FP8_MAX * s).all()
+
+
+def test_quantize_scale_is_per_row():
+    w = torch.zeros(4, 8)
+    w[0, 0] = 448.0  # row 0 amax exactly FP8_MAX -> scale 1
+    w[1, 3] = 44.8
+    q, s = quantize_fp8_rowwise(w)
+    assert torch.isclose(s[0, 0], torch.tensor(1.0))
+    assert torch.isclose(s[1, 0], torch.tensor(0.1), rtol=1e-3)
+    assert float(q[0, 0].float()) == 448.0
+
+
+def test_fp8_linear_close_to_bf16_linear():
+    torch.manual_seed(1)
+    lin = torch.nn.Linear(128, 96, bias=False, dtype=torch.bfloat16)
+    with torch.no_grad():
+        lin.weight.uniform_(-0.05, 0.05)
+    qlin = Fp8Linear(lin)
+    x = torch.randn(32, 128, dtype=torch.bfloat16)
+    y_ref = lin(x).float()
+    y_q = qlin(x).float()
+    # weight + activation quant noise: relative Frobenius error ~1-3%
+    rel = (y_q - y_ref).norm() / y_ref.norm()
+    assert rel < 0.08, rel
+    # shape/dtype contract
+    assert y_q.shape == (32, 96)
+    assert qlin(x).dtype == torch.bfloat16
+
+
+def test_fp8_linear_3d_input():
+    lin = torch.nn.Linear(16, 8, bias=False, dtype=torch.bfloat16)
+    q = Fp8Linear(lin)
+    x = torch.randn(2, 5, 16, dtype=torch.bfloat16)
+    assert q(x).shape == (2, 5, 8)
+
+
+def test_convert_llama_model_counts_and_runs():
+    from llm_instance_gateway_amd.models.config import get_model_config
+    from llm_instance_gateway_amd.models.llama import LlamaForCausalLM
+
+    mc = get_model_config("tiny")
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(mc, device="cpu", dtype=torch.bfloat16)
+    n = convert_linears_to_fp8(model)
+    assert n == 4 * mc.num_layers
+    # projections converted, everything else untouched
+    l0 = model.layers[0]
+    assert isinstance(l0.self_attn.qkv_proj, Fp8Linear)
+    assert isinstance(l0.mlp.down_proj, Fp8Linear)
+    assert isinstance(model.embed_tokens, torch.nn.Embedding)
+
+
+def test_engine_fp8_generates_tokens():
+    """End-to-end: a tiny engine with quant='fp8' decodes valid tokens on
+    CPU (dequantized-reference path)."""
+    from llm_instance_gateway_amd.engine import (
+        EngineConfig,
+        LLMEngine,
+        SamplingParams,
+    )
+    from llm_instance_gateway_amd.models.config import get_model_config
+
+    mc = get_model_config("tiny")
+    cfg = EngineConfig(
+        model=mc,
+        num_kv_blocks=64,
+        max_num_seqs=4,
+        max_num_batched_tokens=128,
+        max_model_len=128,
+        quant="fp8",
+        pad_batch_multiple=0,
+    )
+    torch.manual_seed(7)
+    eng = LLMEngine(cfg, device="cpu")
+    seq = eng.add_request([5, 6, 7, 8], SamplingParams(max_new_tokens=8))
+    for _ in range(16):
+        eng.step()
+        if seq.is_finished():
+            break
+    assert seq.is_finished()
+    assert len(seq.output_token_ids) == 8
+    assert all(0 <= t < mc.vocab_size for t in seq.output_token_ids)
+
+
+def test_fp8_rejects_moe():
+    from llm_instance_gateway_amd.engine import EngineConfig, LLMEngine
+    from llm_instance_gateway_amd.models.config import get_model_config
+
+    mc = get_model_config("tiny-moe")
+    cfg = EngineConfig(
+        model=mc, num_kv_blocks=16, max_num_seqs=2,
+        max_num_batched_tokens=64, max_model_len=64, quant="fp8",
+    )
+    with pytest.raises(ValueError, match="dense"):
+        LLMEngine(cfg, device="cpu")
