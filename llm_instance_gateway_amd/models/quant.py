@@ -23,6 +23,16 @@ reported separately and marked as such.  Reference analogue: vLLM's
 --quantization fp8 serving mode on the fleet the reference routes to
 (SURVEY.md §2.3); implementation is original and MI355X-native.
 
+MEASURED STATUS (r02, 1x MI355X): numerics validated on gfx950
+(tests/test_quant_gpu.py, 4/4) but a NON-win as implemented — 69.4 vs
+74.6 req/s at conc=384 and 27.1 vs 31.6 at conc=64 (gpurun_out/
+bench_fp8.log, bench_c64_*.log): the dynamic activation-quant
+elementwise passes (amax reduce + scale + cast per projection, ~4
+launches each) cost more than the halved weight traffic returns.
+ops.quant_fp8_rowwise fuses the quant to ONE kernel; remaining path to a
+win is the ROADMAP's hybrid dispatch (bf16 eager prefill + fp8 graphed
+decode) or a fused quant+GEMM epilogue.
+
 CPU fallback (tests): the same quantized values are dequantized and the
 matmul runs in fp32 — numerics match the GPU path up to accumulation
 order, so CPU tests pin the quantization semantics.
@@ -68,6 +78,10 @@ class Fp8Linear(nn.Module):
         self._out_dtype = w.dtype
 
     def _quant_input(self, x2: torch.Tensor):
+        if x2.device.type == "cuda":
+            from .. import ops
+
+            return ops.quant_fp8_rowwise(x2)  # ONE fused kernel
         amax = x2.abs().amax(dim=-1, keepdim=True).float().clamp_(min=1e-6)
         xs = amax / FP8_MAX
         xq = (x2 / xs).clamp_(-FP8_MAX, FP8_MAX).to(FP8_DTYPE)

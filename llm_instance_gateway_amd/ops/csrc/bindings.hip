@@ -58,6 +58,8 @@ void lig_sgmv_expand_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             hipStream_t);
 void lig_skinny_gemm_launch(uint16_t*, const uint16_t*, const uint16_t*,
                             int, int, int, hipStream_t);
+void lig_quant_fp8_rowwise_launch(uint8_t*, float*, const uint16_t*, int,
+                                  int, hipStream_t);
 void lig_sgmv_shrink_units_launch(uint16_t*, const uint16_t*, const uint16_t*,
                                   const int*, const int*, const int*,
                                   const int*, int, int, int, hipStream_t);
@@ -378,6 +380,23 @@ void sgmv_expand(torch::Tensor y, torch::Tensor u, torch::Tensor b_stack,
                          T >= 256, as_stream(stream));
 }
 
+void quant_fp8_rowwise(torch::Tensor xq, torch::Tensor xs, torch::Tensor x,
+                       int64_t stream) {
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  TORCH_CHECK(xq.element_size() == 1, "xq must be an 8-bit (fp8) tensor");
+  TORCH_CHECK(xs.scalar_type() == torch::kFloat32, "xs must be fp32");
+  TORCH_CHECK(x.dim() == 2 && xq.sizes() == x.sizes(), "shape mismatch");
+  const int T = x.size(0);
+  const int K = x.size(1);
+  TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+  TORCH_CHECK(xs.numel() == T, "xs must have one scale per row");
+  TORCH_CHECK(x.is_contiguous() && xq.is_contiguous() && xs.is_contiguous());
+  TORCH_CHECK(xq.is_cuda() && xs.is_cuda());
+  lig_quant_fp8_rowwise_launch(reinterpret_cast<uint8_t*>(xq.data_ptr()),
+                               xs.data_ptr<float>(), bf16_cptr(x), T, K,
+                               as_stream(stream));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -396,6 +415,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm);
   m.def("sgmv_apply_tiled", &sgmv_apply_tiled);
   m.def("sgmv_expand", &sgmv_expand);
+  m.def("quant_fp8_rowwise", &quant_fp8_rowwise);
   m.def("mfma_probe", [](torch::Tensor out, torch::Tensor a, torch::Tensor b,
                          int64_t stream) {
     TORCH_CHECK(out.scalar_type() == torch::kFloat32);

@@ -13,6 +13,27 @@ if not torch.cuda.is_available():  # collected on CPU boxes; never run
 from llm_instance_gateway_amd.models.quant import Fp8Linear
 
 
+@pytest.mark.parametrize("T,K", [(1, 4096), (384, 4096), (777, 14336)])
+def test_fused_quant_kernel_matches_torch(T, K):
+    from llm_instance_gateway_amd import ops
+    from llm_instance_gateway_amd.models.quant import FP8_MAX
+
+    torch.manual_seed(0)
+    x = (torch.randn(T, K, dtype=torch.bfloat16) * 0.3).cuda()
+    xq, xs = ops.quant_fp8_rowwise(x)
+    assert xq.dtype == torch.float8_e4m3fn and xs.shape == (T, 1)
+    # reference: torch elementwise path on the same device
+    amax = x.abs().amax(dim=-1, keepdim=True).float().clamp_(min=1e-6)
+    rs = amax / FP8_MAX
+    rq = (x / rs).clamp_(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+    assert torch.allclose(xs, rs, rtol=1e-6)
+    # e4m3 values: identical up to an occasional 1-ulp rounding-boundary flip
+    # from inv-scale multiply vs scale divide
+    diff = (xq.float() - rq.float()).abs()
+    denom = rq.float().abs().clamp(min=1e-3)
+    assert (diff / denom > 0.15).float().mean() < 1e-3
+
+
 @pytest.mark.parametrize("M,K,N", [(1, 4096, 6144), (256, 4096, 6144), (384, 14336, 4096)])
 def test_scaled_mm_matches_dequant_reference(M, K, N):
     torch.manual_seed(0)
