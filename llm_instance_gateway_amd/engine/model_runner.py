@@ -332,8 +332,22 @@ class ModelRunner:
         sampled: Dict[int, int] = {}
         if sample_seqs:
             with gpu_range("sample"):
-                idx = torch.tensor(sample_idx, dtype=torch.int64, device=self.device)
+                # Bucket the logits GEMM's M (ROADMAP r02 #4): sample counts
+                # vary 1..max_num_seqs per mixed step — pad the gather index
+                # to a multiple of 64 (repeat row 0) so the 128k-vocab GEMM
+                # sees a handful of stable shapes; pad rows sliced off.
+                M = len(sample_idx)
+                pad_m = 0
+                if self.device.type == "cuda" and self.config.pad_batch_multiple:
+                    pad_m = (-M) % 64
+                idx = torch.tensor(
+                    sample_idx + [sample_idx[0]] * pad_m,
+                    dtype=torch.int64,
+                    device=self.device,
+                )
                 logits = self.model.compute_logits(hidden, idx)
+                if pad_m:
+                    logits = logits[:M]
                 tokens = sample_tokens(logits, sample_seqs)
             for seq, tok in zip(sample_seqs, tokens):
                 sampled[seq.seq_id] = tok

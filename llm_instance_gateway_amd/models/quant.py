@@ -23,15 +23,16 @@ reported separately and marked as such.  Reference analogue: vLLM's
 --quantization fp8 serving mode on the fleet the reference routes to
 (SURVEY.md §2.3); implementation is original and MI355X-native.
 
-MEASURED STATUS (r02, 1x MI355X): numerics validated on gfx950
-(tests/test_quant_gpu.py, 4/4) but a NON-win as implemented — 69.4 vs
-74.6 req/s at conc=384 and 27.1 vs 31.6 at conc=64 (gpurun_out/
-bench_fp8.log, bench_c64_*.log): the dynamic activation-quant
-elementwise passes (amax reduce + scale + cast per projection, ~4
-launches each) cost more than the halved weight traffic returns.
-ops.quant_fp8_rowwise fuses the quant to ONE kernel; remaining path to a
-win is the ROADMAP's hybrid dispatch (bf16 eager prefill + fp8 graphed
-decode) or a fused quant+GEMM epilogue.
+MEASURED STATUS (r02, 1x MI355X, tests/test_quant_gpu.py 7/7 green):
+  * with torch elementwise activation quant (~4 launches + 3 passes over
+    x per projection): a NON-win — 69.4 vs 74.6 req/s at conc=384
+    (gpurun_out/bench_fp8.log);
+  * with the fused ONE-launch quant kernel (ops.quant_fp8_rowwise,
+    ops/csrc/quant.hip): **84.97 vs 74.6 req/s at conc=384 (+14%)** and
+    35.7 vs 31.6 at conc=64 (+13%) — bench_fp8_fused.log,
+    bench_c64_fp8_fused.log.  The quant overhead was the entire penalty.
+The headline benchmark still reports bf16 (BASELINE's dtype); fp8 runs
+carry dtype="fp8" in the bench JSON.
 
 CPU fallback (tests): the same quantized values are dequantized and the
 matmul runs in fp32 — numerics match the GPU path up to accumulation
