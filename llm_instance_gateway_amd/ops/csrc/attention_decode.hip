@@ -27,6 +27,19 @@
 
 #include "common.h"
 
+typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
+
+// h-th bf16 pair of a packed u16x8 (folds to subregister access once the
+// caller's loop is unrolled)
+static __device__ __forceinline__ bf16x2_t bf16pair(const u16x8& v, int h) {
+  union {
+    u16x8 w;
+    bf16x2_t p[4];
+  } u;
+  u.w = v;
+  return u.p[h];
+}
+
 // SPLIT=true: flash-decoding — blockIdx.z selects a context slice; the
 // un-normalized partial (oacc, m, l) goes to scratch fp32
 // [N, Hq, nsplits, D+2] and a combine kernel merges the slices.  Doubles
@@ -36,8 +49,14 @@
 // instantiation with launch_bounds(256) lifts the 1024-thread VGPR ceiling
 // (110 VGPRs -> 4 waves/SIMD at LB=1024) so the scheduler can trade
 // registers for occupancy per the real block size.
-template <int HEAD_DIM, bool SPLIT = false, int MAX_THREADS = 1024>
-__global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
+// MIN_BLOCKS (launch_bounds 2nd arg): workgroups the compiler must fit per
+// CU.  The 256-thread (GQA group 4) instantiations pin 5 -> 5 waves/SIMD,
+// forcing VGPRs <= 96 (the allocator otherwise settles at 98 = 4 waves;
+// the packed-bf16 dot + fused V-convert brought it down from 110).
+template <int HEAD_DIM, bool SPLIT = false, int MAX_THREADS = 1024,
+          int MIN_BLOCKS = 1>
+__global__ __launch_bounds__(MAX_THREADS, MIN_BLOCKS) void
+paged_attention_decode_kernel(
     uint16_t* __restrict__ out,            // [N, Hq, D]
     const uint16_t* __restrict__ q,        // [N, Hq, D]
     const uint16_t* __restrict__ k_cache,  // [B, Hkv, BS, D]
@@ -63,20 +82,22 @@ __global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
   const int tok = lane >> 3;     // 0..7: token within the 8-token iteration
   const int dchunk = lane & 7;   // 0..7: which head_dim/8 slice
 
-  // q fragment -> f32 registers
-  float qreg[DL];
+  // q fragment: PACKED bf16 for DL>=8 (v_dot2c_f32_bf16 consumes bf16
+  // pairs directly — halves the q registers vs an f32 copy and removes the
+  // per-iteration K conversion buffer; the kernel is VGPR-occupancy-bound:
+  // 106 VGPRs = 4 waves/SIMD, <=96 buys a 5th); f32 for the small DL<8
+  // instantiations (HEAD_DIM 16, test-only shapes).
+  constexpr int NQ = DL >= 8 ? DL / 8 : 1;
+  u16x8 qpk[NQ];
+  float qreg[DL < 8 ? DL : 1];
   {
     const uint16_t* qp =
         q + ((size_t)seq * num_q_heads + qh) * HEAD_DIM + dchunk * DL;
+    if constexpr (DL >= 8) {
 #pragma unroll
-    for (int i = 0; i < DL; i += 8) {
-      if constexpr (DL >= 8) {
-        f32x8 v8 = bf16x8_to_f32x8(*reinterpret_cast<const u16x8*>(qp + i));
-#pragma unroll
-        for (int j = 0; j < 8; ++j) qreg[i + j] = v8[j];
-      }
-    }
-    if constexpr (DL < 8) {
+      for (int i = 0; i < DL / 8; ++i)
+        qpk[i] = *reinterpret_cast<const u16x8*>(qp + i * 8);
+    } else {
 #pragma unroll
       for (int i = 0; i < DL; ++i) qreg[i] = bf16_to_f32(qp[i]);
     }
@@ -138,19 +159,21 @@ __global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
 #define STEP_BODY(KBUF, VBUF, VALID)                                         \
   do {                                                                       \
     const bool valid = (VALID);                                              \
-    float kf[DL];                                                            \
+    float s = 0.0f;                                                          \
     if constexpr (DL >= 8) {                                                 \
-      _Pragma("unroll") for (int i = 0; i < NV; ++i) {                       \
-        f32x8 v8 = bf16x8_to_f32x8((KBUF)[i]);                               \
-        _Pragma("unroll") for (int j = 0; j < 8; ++j)                        \
-            kf[i * 8 + j] = valid ? v8[j] : 0.0f;                            \
+      /* packed-bf16 dot: no f32 K buffer; an invalid lane's stale K     */ \
+      /* registers feed a dot whose result is discarded by the select    */ \
+      /* below (s := -INF), so no zeroing pass is needed.                */ \
+      _Pragma("unroll") for (int i = 0; i < NQ; ++i) {                       \
+        _Pragma("unroll") for (int h = 0; h < 4; ++h) {                      \
+          s = __builtin_amdgcn_fdot2_f32_bf16(                               \
+              bf16pair(qpk[i], h), bf16pair((KBUF)[i], h), s, false);        \
+        }                                                                    \
       }                                                                      \
     } else {                                                                 \
       _Pragma("unroll") for (int i = 0; i < DL; ++i)                         \
-          kf[i] = valid ? bf16_to_f32((KBUF)[0][i]) : 0.0f;                  \
+          s += qreg[i] * (valid ? bf16_to_f32((KBUF)[0][i]) : 0.0f);         \
     }                                                                        \
-    float s = 0.0f;                                                          \
-    _Pragma("unroll") for (int i = 0; i < DL; ++i) s += qreg[i] * kf[i];     \
     s = group_reduce_sum<8>(s);                                              \
     s = valid ? s * scale : -INFINITY;                                       \
     /* PER-TOKEN-STREAM online softmax: the 8 lanes of this token group    */\
@@ -170,19 +193,21 @@ __global__ __launch_bounds__(MAX_THREADS) void paged_attention_decode_kernel(
       m = new_m;                                                             \
     }                                                                        \
     l = l * alpha + p;                                                       \
-    float vf[DL];                                                            \
+    /* V accumulate with inline conversion (no vf[DL] buffer — VGPRs).   */ \
+    /* Invalid lanes force the converted value to 0 so stale V registers */ \
+    /* cannot inject NaN through p * v (p is 0 there, but 0*NaN = NaN).  */ \
     if constexpr (DL >= 8) {                                                 \
       _Pragma("unroll") for (int i = 0; i < NV; ++i) {                       \
         f32x8 v8 = bf16x8_to_f32x8((VBUF)[i]);                               \
         _Pragma("unroll") for (int j = 0; j < 8; ++j)                        \
-            vf[i * 8 + j] = valid ? v8[j] : 0.0f;                            \
+            oacc[i * 8 + j] =                                                \
+                oacc[i * 8 + j] * alpha + p * (valid ? v8[j] : 0.0f);        \
       }                                                                      \
     } else {                                                                 \
       _Pragma("unroll") for (int i = 0; i < DL; ++i)                         \
-          vf[i] = valid ? bf16_to_f32((VBUF)[0][i]) : 0.0f;                  \
+          oacc[i] = oacc[i] * alpha +                                        \
+                    p * (valid ? bf16_to_f32((VBUF)[0][i]) : 0.0f);          \
     }                                                                        \
-    _Pragma("unroll") for (int i = 0; i < DL; ++i)                           \
-        oacc[i] = oacc[i] * alpha + p * vf[i];                               \
   } while (0)
 
   ISSUE_LOAD(kraw0, vraw0, valid0, it0);
@@ -296,7 +321,7 @@ extern "C" void lig_paged_attention_decode_split_launch(
 #define LAUNCH_S(HD)                                                        \
   do {                                                                      \
     if (group <= 4)                                                         \
-      hipLaunchKernelGGL((paged_attention_decode_kernel<HD, true, 256>),    \
+      hipLaunchKernelGGL((paged_attention_decode_kernel<HD, true, 256, 5>),    \
                          grid, block, 0, stream, out, q, k_cache, v_cache,  \
                          block_tables, context_lens, scale, num_q_heads,    \
                          num_kv_heads, block_size, max_blocks, scratch,     \
@@ -332,7 +357,7 @@ extern "C" void lig_paged_attention_decode_launch(
 #define LAUNCH(HD)                                                          \
   do {                                                                      \
     if (group <= 4)                                                         \
-      hipLaunchKernelGGL((paged_attention_decode_kernel<HD, false, 256>),   \
+      hipLaunchKernelGGL((paged_attention_decode_kernel<HD, false, 256, 5>),   \
                          grid, block, 0, stream, out, q, k_cache, v_cache,  \
                          block_tables, context_lens, scale, num_q_heads,    \
                          num_kv_heads, block_size, max_blocks);             \
