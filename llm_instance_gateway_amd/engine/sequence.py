@@ -38,9 +38,14 @@ class SamplingParams:
     # OpenAI `stop`: token-id sequences; generation halts when the output
     # tail matches one, and the matched tail is trimmed from the output.
     # Token-id matching equals string matching under the front door's
-    # byte-level codec (utils/tokenizer.py: one byte == one id, context-free);
-    # a subword tokenizer would need decoded-text matching instead.
+    # byte-level codec (utils/tokenizer.py: one byte == one id, context-free).
     stop_sequences: List[List[int]] = field(default_factory=list)
+    # String-level stops (vLLM semantics): with a subword tokenizer a stop
+    # string can tokenize differently in generation context, so the engine
+    # ALSO matches stops on the decoded output tail (utils.tokenizer.decode
+    # of a bounded suffix).  The front door trims the stop text from the
+    # decoded response; token counts are not trimmed on this path.
+    stop_texts: List[str] = field(default_factory=list)
 
 
 class Sequence:
@@ -136,6 +141,18 @@ class Sequence:
                 del self.output_token_ids[-n:]
                 self.finish_reason = "stop"
                 return True
+        if s.stop_texts and self.output_token_ids:
+            from ..utils import tokenizer as _tok
+
+            # A stop of L chars spans at most L tokens (every token decodes
+            # to >= 1 char); +8 slack for multi-char boundary tokens.
+            longest = max(len(t) for t in s.stop_texts)
+            ntail = min(self.num_output_tokens, longest + 8)
+            tail = _tok.decode(self.output_token_ids[-ntail:])
+            for t in s.stop_texts:
+                if t and t in tail:
+                    self.finish_reason = "stop"
+                    return True
         return False
 
     def preempt(self) -> None:

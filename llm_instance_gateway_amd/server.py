@@ -18,6 +18,7 @@ config-file path that replaces the K8s API.
 from __future__ import annotations
 
 import argparse
+import os
 import logging
 import signal
 import sys
@@ -108,6 +109,9 @@ def main(argv: Optional[List[str]] = None) -> int:
     if args.tokenizer:
         if _tok.load_backend(args.tokenizer):
             log.info("front-door tokenizer: %s", _tok.backend_name())
+            # engine workers (spawned later) load the same backend for
+            # decoded-text stop matching
+            os.environ["LIG_TOKENIZER"] = args.tokenizer
         else:
             log.warning("could not load tokenizer %s; using byte codec", args.tokenizer)
 

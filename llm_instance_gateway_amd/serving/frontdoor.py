@@ -154,6 +154,10 @@ class Gateway:
             critical=critical,
             stream=req.stream,
             stop_sequences=[tokenizer.encode(t, add_bos=False) for t in stops if t],
+            # string-level matching too (vLLM semantics): with a subword
+            # tokenizer the stop can tokenize differently in generation
+            # context and the token-id match above would miss it
+            stop_texts=[t for t in stops if t],
         )
         self.spans.event(preq.request_id, "arrived", model=req.model,
                          target=target, critical=critical)
@@ -177,6 +181,21 @@ class Gateway:
         self.spans.event(fin.request_id, "finished", engine=fin.engine_name,
                          ttft=fin.ttft, e2e=fin.e2e,
                          output_tokens=len(fin.output_token_ids))
+
+
+def _trim_stop_text(text: str, req_stop) -> str:
+    """OpenAI semantics: the matched stop text (and anything after it) is
+    not returned.  Applied to DECODED text when the engine finished on a
+    stop — covers the decoded-text match path, where output token ids are
+    not trimmed (engine/sequence.py stop_texts)."""
+    stops = [req_stop] if isinstance(req_stop, str) else (req_stop or [])
+    cut = len(text)
+    for s in stops:
+        if s:
+            i = text.find(s)
+            if i >= 0:
+                cut = min(cut, i)
+    return text[:cut]
 
 
 def _usage(prompt_len: int, completion_len: int) -> Dict[str, int]:
@@ -223,10 +242,15 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
             # Hold back a tail of len(longest stop) tokens until the stop
             # check clears them; at stream end flush from the engine's
             # trimmed output so matched stop tokens are dropped.
-            stop_lists = [tokenizer.encode(s, add_bos=False)
-                          for s in ([req.stop] if isinstance(req.stop, str)
-                                    else (req.stop or [])) if s]
-            holdback = max((len(s) for s in stop_lists), default=0)
+            stops_s = [s for s in ([req.stop] if isinstance(req.stop, str)
+                                   else (req.stop or [])) if s]
+            # a stop of L chars spans at most L generated tokens, which can
+            # exceed its standalone encoding length under a subword tokenizer
+            holdback = max(
+                [len(tokenizer.encode(s, add_bos=False)) for s in stops_s]
+                + [len(s) for s in stops_s],
+                default=0,
+            )
 
             async def sse() -> AsyncIterator[str]:
                 loop = asyncio.get_running_loop()
@@ -264,9 +288,22 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
                 gw.record_finished(ticket, req.model)
                 fin = ticket.result
                 if fin is not None:
-                    # flush survivors of the stop trim (held-back tail)
-                    for tok in fin.output_token_ids[sent:]:
-                        yield chunk_for(tok)
+                    # flush survivors of the stop trim (held-back tail) as one
+                    # text chunk, with any decoded-text-matched stop removed
+                    tail_text = tokenizer.decode(fin.output_token_ids[sent:])
+                    if fin.finish_reason == "stop":
+                        tail_text = _trim_stop_text(tail_text, req.stop)
+                    if tail_text:
+                        yield (
+                            "data: "
+                            + json.dumps({
+                                "id": rid, "object": "text_completion",
+                                "created": created, "model": req.model,
+                                "choices": [{"index": 0, "text": tail_text,
+                                             "finish_reason": None}],
+                            })
+                            + "\n\n"
+                        )
                 final = {
                     "id": rid,
                     "object": "text_completion",
@@ -323,10 +360,13 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
             gw.record_finished(t, req.model)
             fin_i = t.result
             total_completion += len(fin_i.output_token_ids)
+            out_text = tokenizer.decode(fin_i.output_token_ids)
+            if fin_i.finish_reason == "stop":
+                out_text = _trim_stop_text(out_text, req.stop)
             choices.append(
                 {
                     "index": i,
-                    "text": echo_text + tokenizer.decode(fin_i.output_token_ids),
+                    "text": echo_text + out_text,
                     "token_ids": fin_i.output_token_ids,
                     "finish_reason": fin_i.finish_reason,
                 }

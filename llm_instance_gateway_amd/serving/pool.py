@@ -70,6 +70,7 @@ class PoolRequest:
     critical: bool = False
     stream: bool = False
     stop_sequences: List[List[int]] = field(default_factory=list)
+    stop_texts: List[str] = field(default_factory=list)
     arrival_time: float = field(default_factory=time.monotonic)
 
 
@@ -151,6 +152,14 @@ def _worker_main(
         from ..utils.tunableop import maybe_enable_tunableop
 
         maybe_enable_tunableop()
+    # Worker processes need the same tokenizer backend as the front door for
+    # decoded-text stop matching (server.py exports LIG_TOKENIZER; spawn
+    # children inherit the env).
+    tok_path = os.environ.get("LIG_TOKENIZER", "")
+    if tok_path:
+        from ..utils import tokenizer as _worker_tok
+
+        _worker_tok.load_backend(tok_path)
     tp = None
     if tp_size > 1:
         from ..parallel import TPGroup, init_distributed
@@ -230,6 +239,7 @@ def _worker_main(
                         temperature=r.temperature,
                         top_p=r.top_p,
                         stop_sequences=r.stop_sequences,
+                        stop_texts=r.stop_texts,
                     ),
                     lora_name=r.lora_name,
                     request_id=r.request_id,
