@@ -105,7 +105,10 @@ def parse_config(raw: Dict[str, Any]) -> GatewayConfig:
             base_model=str(p.get("base_model", "")),
             tensor_parallel=int(p.get("tensor_parallel", 1)),
             devices=[str(d) for d in p.get("devices", [])],
+            quant=str(p.get("quant", "")),
         )
+        if cfg.pool.quant not in ("", "fp8"):
+            raise ConfigError(f"pool.quant must be '' or 'fp8', got {cfg.pool.quant!r}")
 
     for m in raw.get("models", []) or []:
         if not isinstance(m, dict) or "model_name" not in m:

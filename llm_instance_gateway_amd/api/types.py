@@ -103,6 +103,9 @@ class InferencePool:
     # Explicit device names (overrides gpu_ids; lets CPU deployments/tests
     # drive runtime membership reconciliation the same way).
     devices: List[str] = field(default_factory=list)
+    # Engine quantization mode: "" (bf16) or "fp8" (e4m3 projection GEMMs,
+    # models/quant.py — vLLM --quantization fp8 analogue; dense models only).
+    quant: str = ""
 
 
 def random_weighted_draw(model: InferenceModel, seed: int = 0) -> str:

@@ -51,6 +51,8 @@ def build_pool(args, cfg) -> "EnginePool":
             num_kv_blocks=256, max_num_seqs=16,
             max_num_batched_tokens=512, max_model_len=512,
         )
+    if cfg and cfg.pool.quant:
+        engine_kwargs["quant"] = cfg.pool.quant
     tp = cfg.pool.tensor_parallel if cfg else 1
     if cfg and tp > 1 and not args.devices:
         # TP pools: chunk devices into TP groups, one engine per group —

@@ -30,6 +30,7 @@ pool:
   name: mi355x-pool
   base_model: tiny
   gpu_ids: [0, 1]
+  quant: fp8
 models:
   - model_name: chat
     criticality: Critical
@@ -65,6 +66,7 @@ def test_parse_config_full():
     cfg = parse_config(yaml.safe_load(CFG_YAML))
     assert cfg.pool.name == "mi355x-pool"
     assert cfg.pool.gpu_ids == [0, 1]
+    assert cfg.pool.quant == "fp8"
     assert [m.model_name for m in cfg.models] == ["chat", "batch", "other-pool-model"]
     assert cfg.models[0].criticality == Criticality.CRITICAL
     assert cfg.models[0].target_models[1].weight == 40
@@ -76,6 +78,11 @@ def test_parse_config_full():
     assert cfg.queueing_threshold_lora == 25
     assert cfg.refresh_metrics_interval_s == 0.01
     assert cfg.port == 9099
+
+
+def test_parse_config_rejects_bad_quant():
+    with pytest.raises(ConfigError, match="quant"):
+        parse_config(yaml.safe_load("pool: {name: p, quant: int4}"))
 
 
 @pytest.mark.parametrize(
