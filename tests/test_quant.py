@@ -113,6 +113,45 @@ def test_engine_fp8_generates_tokens():
     assert all(0 <= t < mc.vocab_size for t in seq.output_token_ids)
 
 
+def test_engine_fp8_with_lora_adapters():
+    """LoRA rides on top of fp8 base GEMMs unchanged: SGMV applies bf16
+    deltas to the (bf16) GEMM output, so the adapter path must keep
+    working when the projections are Fp8Linear."""
+    from llm_instance_gateway_amd.engine import (
+        EngineConfig,
+        LLMEngine,
+        SamplingParams,
+    )
+    from llm_instance_gateway_amd.lora.manager import AdapterManager
+    from llm_instance_gateway_amd.models.config import get_model_config
+
+    mc = get_model_config("tiny")
+    cfg = EngineConfig(
+        model=mc,
+        num_kv_blocks=64,
+        max_num_seqs=4,
+        max_num_batched_tokens=128,
+        max_model_len=128,
+        quant="fp8",
+        max_loras=2,
+        pad_batch_multiple=0,
+    )
+    torch.manual_seed(11)
+    eng = LLMEngine(cfg, device="cpu")
+    mgr = AdapterManager(eng, max_slots=2, rank=8)
+    mgr.register_random("a1", seed=1)
+    mgr.prepare(["a1"])
+    base = eng.add_request([5, 6, 7], SamplingParams(max_new_tokens=6))
+    lora = eng.add_request([5, 6, 7], SamplingParams(max_new_tokens=6), lora_name="a1")
+    for _ in range(24):
+        eng.step()
+        if base.is_finished() and lora.is_finished():
+            break
+    assert base.is_finished() and lora.is_finished()
+    # the adapter delta must actually change the output
+    assert base.output_token_ids != lora.output_token_ids
+
+
 def test_fp8_rejects_moe():
     from llm_instance_gateway_amd.engine import EngineConfig, LLMEngine
     from llm_instance_gateway_amd.models.config import get_model_config
