@@ -156,8 +156,10 @@ class Gateway:
             stop_sequences=[tokenizer.encode(t, add_bos=False) for t in stops if t],
             # string-level matching too (vLLM semantics): with a subword
             # tokenizer the stop can tokenize differently in generation
-            # context and the token-id match above would miss it
-            stop_texts=[t for t in stops if t],
+            # context and the token-id match above would miss it.  Under the
+            # byte codec token matching IS string matching, so skip the
+            # per-token decoded-tail checks there.
+            stop_texts=[t for t in stops if t] if tokenizer.has_backend() else [],
         )
         self.spans.event(preq.request_id, "arrived", model=req.model,
                          target=target, critical=critical)

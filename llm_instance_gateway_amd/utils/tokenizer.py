@@ -50,6 +50,13 @@ def backend_name() -> str:
     return type(_BACKEND).__name__ if _BACKEND is not None else "byte-codec"
 
 
+def has_backend() -> bool:
+    """True when a real (subword) tokenizer is installed.  Under the byte
+    codec token-id stop matching is exactly string matching (1 byte = 1 id,
+    context-free), so decoded-text stop checks are redundant."""
+    return _BACKEND is not None
+
+
 def encode(text: str, add_bos: bool = True) -> List[int]:
     if _BACKEND is not None:
         return list(_BACKEND.encode(text).ids)
