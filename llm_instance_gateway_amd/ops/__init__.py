@@ -252,10 +252,12 @@ def quant_fp8_rowwise(x: torch.Tensor):
         xs = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
         _EXT.quant_fp8_rowwise(xq, xs, x, _stream())
         return xq, xs.unsqueeze(1)
-    from ..models.quant import quantize_fp8_rowwise
-
-    q, s = quantize_fp8_rowwise(x)
-    return q, s
+    # CPU reference of the KERNEL's exact semantics (amax clamp 1e-6,
+    # matching models/quant.Fp8Linear._quant_input's activation path)
+    amax = x.abs().amax(dim=-1, keepdim=True).float().clamp_(min=1e-6)
+    xs = amax / 448.0
+    xq = (x / xs).clamp_(-448.0, 448.0).to(torch.float8_e4m3fn)
+    return xq, xs
 
 
 def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
