@@ -42,6 +42,12 @@ def sample_tokens(logits: torch.Tensor, seqs: List[Sequence]) -> List[int]:
             probs = probs.masked_fill(~mask, 0.0)
             probs = probs / probs.sum()
         gen = torch.Generator(device=probs.device)
-        gen.manual_seed((sp.seed or 0) * 1_000_003 + seq.seq_id * 7919 + seq.num_output_tokens)
+        if sp.seed:
+            # OpenAI `seed` contract: the same request + seed reproduces the
+            # same tokens, independent of engine-local sequence numbering
+            gen.manual_seed(sp.seed * 1_000_003 + seq.num_output_tokens)
+        else:
+            # unseeded: decorrelate concurrent sequences via the seq id
+            gen.manual_seed(seq.seq_id * 7919 + seq.num_output_tokens)
         out.append(int(torch.multinomial(probs, 1, generator=gen)))
     return out

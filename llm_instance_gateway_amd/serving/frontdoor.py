@@ -57,6 +57,7 @@ class CompletionRequest(BaseModel):
     stop: Optional[Union[str, List[str]]] = None
     n: int = 1            # number of choices (non-streaming only)
     echo: bool = False    # prepend the prompt text to each choice
+    seed: Optional[int] = None  # deterministic sampling (OpenAI seed field)
     # non-standard extensions (the bench/tests use these)
     lora: Optional[str] = None  # explicit adapter override
     request_id: Optional[str] = None
@@ -150,6 +151,7 @@ class Gateway:
             max_new_tokens=req.max_tokens,
             temperature=req.temperature,
             top_p=req.top_p,
+            seed=(req.seed if req.seed is not None else 0) + choice_idx,
             lora_name=lora,
             critical=critical,
             stream=req.stream,

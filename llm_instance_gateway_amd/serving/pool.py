@@ -66,6 +66,7 @@ class PoolRequest:
     max_new_tokens: int = 128
     temperature: float = 0.0
     top_p: float = 1.0
+    seed: int = 0
     lora_name: str = ""
     critical: bool = False
     stream: bool = False
@@ -238,6 +239,7 @@ def _worker_main(
                         max_new_tokens=r.max_new_tokens,
                         temperature=r.temperature,
                         top_p=r.top_p,
+                        seed=r.seed,
                         stop_sequences=r.stop_sequences,
                         stop_texts=r.stop_texts,
                     ),

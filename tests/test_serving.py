@@ -330,6 +330,20 @@ def test_unknown_adapter_404(client):
     assert "unknown LoRA adapter" in r.json()["error"]
 
 
+def test_seeded_sampling_deterministic(client):
+    """OpenAI `seed`: same seed + temperature>0 reproduces the choice;
+    different seeds diverge (with overwhelming probability on 128k vocab)."""
+    body = {"model": "my-model", "prompt": "seeded", "max_tokens": 8,
+            "temperature": 0.8, "seed": 42}
+    t1 = client.post("/v1/completions", json=body).json()["choices"][0]["token_ids"]
+    t2 = client.post("/v1/completions", json=body).json()["choices"][0]["token_ids"]
+    assert t1 == t2
+    t3 = client.post(
+        "/v1/completions", json={**body, "seed": 43}
+    ).json()["choices"][0]["token_ids"]
+    assert t3 != t1
+
+
 def test_concurrent_adapter_churn_through_frontdoor(client, pool):
     """Config #5's churn at the serving layer: concurrent requests over more
     adapters than slots (max_loras=4/engine) through the real front door."""
