@@ -46,6 +46,15 @@ class SamplingParams:
     # of a bounded suffix).  The front door trims the stop text from the
     # decoded response; token counts are not trimmed on this path.
     stop_texts: List[str] = field(default_factory=list)
+    # OpenAI `logprobs`: when > 0, record the chosen token's logprob and the
+    # top-N alternatives per step (engine/sampling.py fills
+    # Sequence.output_logprobs).  0 = off (no cost on the sampling path).
+    logprobs: int = 0
+    # vLLM-style penalties, applied to the logits row before sampling;
+    # neutral values are skipped entirely.
+    presence_penalty: float = 0.0   # subtract once per already-present token
+    frequency_penalty: float = 0.0  # subtract per occurrence
+    repetition_penalty: float = 1.0  # divide positive / multiply negative
 
 
 class Sequence:
@@ -62,6 +71,9 @@ class Sequence:
         self.request_id = request_id or f"req-{self.seq_id}"
         self.prompt_token_ids = list(prompt_token_ids)
         self.output_token_ids: List[int] = []
+        # per-step logprob records (only when sampling.logprobs > 0):
+        # [(chosen_logprob, {top_token_id: logprob, ...}), ...]
+        self.output_logprobs: List = []
         self.sampling = sampling or SamplingParams()
         self.lora_name = lora_name
         self.critical = critical

@@ -58,6 +58,10 @@ class CompletionRequest(BaseModel):
     n: int = 1            # number of choices (non-streaming only)
     echo: bool = False    # prepend the prompt text to each choice
     seed: Optional[int] = None  # deterministic sampling (OpenAI seed field)
+    logprobs: Optional[int] = None      # top-N logprobs per generated token
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0     # vLLM extension
     # non-standard extensions (the bench/tests use these)
     lora: Optional[str] = None  # explicit adapter override
     request_id: Optional[str] = None
@@ -152,6 +156,10 @@ class Gateway:
             temperature=req.temperature,
             top_p=req.top_p,
             seed=(req.seed if req.seed is not None else 0) + choice_idx,
+            logprobs=req.logprobs or 0,
+            presence_penalty=req.presence_penalty,
+            frequency_penalty=req.frequency_penalty,
+            repetition_penalty=req.repetition_penalty,
             lora_name=lora,
             critical=critical,
             stream=req.stream,
@@ -367,14 +375,26 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
             out_text = tokenizer.decode(fin_i.output_token_ids)
             if fin_i.finish_reason == "stop":
                 out_text = _trim_stop_text(out_text, req.stop)
-            choices.append(
-                {
-                    "index": i,
-                    "text": echo_text + out_text,
-                    "token_ids": fin_i.output_token_ids,
-                    "finish_reason": fin_i.finish_reason,
+            choice = {
+                "index": i,
+                "text": echo_text + out_text,
+                "token_ids": fin_i.output_token_ids,
+                "finish_reason": fin_i.finish_reason,
+            }
+            if req.logprobs and fin_i.output_logprobs:
+                choice["logprobs"] = {
+                    "tokens": [
+                        tokenizer.decode([t]) for t in fin_i.output_token_ids
+                    ],
+                    "token_logprobs": [
+                        lp for lp, _ in fin_i.output_logprobs
+                    ],
+                    "top_logprobs": [
+                        {tokenizer.decode([t]): v for t, v in top.items()}
+                        for _, top in fin_i.output_logprobs
+                    ],
                 }
-            )
+            choices.append(choice)
         fin = ticket.result
         body = {
             "id": rid,

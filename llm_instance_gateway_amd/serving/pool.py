@@ -67,6 +67,10 @@ class PoolRequest:
     temperature: float = 0.0
     top_p: float = 1.0
     seed: int = 0
+    logprobs: int = 0
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0
     lora_name: str = ""
     critical: bool = False
     stream: bool = False
@@ -85,6 +89,9 @@ class FinishedRequest:
     tpot: Optional[float]
     e2e: float
     finish_reason: str = "length"
+    # only populated when the request asked for logprobs:
+    # [(chosen_logprob, {top_token_id: logprob}), ...]
+    output_logprobs: List = field(default_factory=list)
 
 
 @dataclass
@@ -240,6 +247,10 @@ def _worker_main(
                         temperature=r.temperature,
                         top_p=r.top_p,
                         seed=r.seed,
+                        logprobs=r.logprobs,
+                        presence_penalty=r.presence_penalty,
+                        frequency_penalty=r.frequency_penalty,
+                        repetition_penalty=r.repetition_penalty,
                         stop_sequences=r.stop_sequences,
                         stop_texts=r.stop_texts,
                     ),
@@ -339,6 +350,7 @@ def _worker_main(
                             tpot=seq.tpot(),
                             e2e=(seq.finish_time or now) - seq.arrival_time,
                             finish_reason=seq.finish_reason,
+                            output_logprobs=list(seq.output_logprobs),
                         )
                     )
                 conn.send(("finished", out))

@@ -330,6 +330,25 @@ def test_unknown_adapter_404(client):
     assert "unknown LoRA adapter" in r.json()["error"]
 
 
+def test_logprobs_response_shape(client):
+    """OpenAI `logprobs`: per-token chosen logprob + top-N alternatives."""
+    r = client.post(
+        "/v1/completions",
+        json={"model": "my-model", "prompt": "lp", "max_tokens": 3,
+              "logprobs": 2},
+    )
+    assert r.status_code == 200
+    ch = r.json()["choices"][0]
+    lp = ch["logprobs"]
+    assert len(lp["token_logprobs"]) == 3
+    assert len(lp["tokens"]) == 3
+    assert all(isinstance(v, float) and v <= 0.0 for v in lp["token_logprobs"])
+    assert all(len(top) == 2 for top in lp["top_logprobs"])
+    # the greedy choice's logprob equals the max of its top alternatives
+    for chosen, top in zip(lp["token_logprobs"], lp["top_logprobs"]):
+        assert abs(chosen - max(top.values())) < 1e-6
+
+
 def test_seeded_sampling_deterministic(client):
     """OpenAI `seed`: same seed + temperature>0 reproduces the choice;
     different seeds diverge (with overwhelming probability on 128k vocab)."""
