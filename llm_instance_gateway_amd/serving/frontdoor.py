@@ -79,6 +79,11 @@ class ChatCompletionRequest(BaseModel):
     temperature: float = 0.0
     top_p: float = 1.0
     stream: bool = False
+    stop: Optional[Union[str, List[str]]] = None
+    seed: Optional[int] = None
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    lora: Optional[str] = None
 
 
 class LoadAdapterRequest(BaseModel):
@@ -422,6 +427,11 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
             temperature=req.temperature,
             top_p=req.top_p,
             stream=False,
+            stop=req.stop,
+            seed=req.seed,
+            presence_penalty=req.presence_penalty,
+            frequency_penalty=req.frequency_penalty,
+            lora=req.lora,
         )
         resp = await completions(creq)
         if isinstance(resp, JSONResponse) and resp.status_code == 200:
