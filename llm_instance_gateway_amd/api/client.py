@@ -58,6 +58,14 @@ class GatewayClient:
         temperature: float = 0.0,
         top_p: float = 1.0,
         lora: Optional[str] = None,
+        stop: Optional[Union[str, Sequence[str]]] = None,
+        seed: Optional[int] = None,
+        logprobs: Optional[int] = None,
+        n: int = 1,
+        echo: bool = False,
+        presence_penalty: float = 0.0,
+        frequency_penalty: float = 0.0,
+        repetition_penalty: float = 1.0,
     ) -> Dict[str, Any]:
         body: Dict[str, Any] = {
             "model": model,
@@ -65,9 +73,20 @@ class GatewayClient:
             "max_tokens": max_tokens,
             "temperature": temperature,
             "top_p": top_p,
+            "n": n,
+            "echo": echo,
+            "presence_penalty": presence_penalty,
+            "frequency_penalty": frequency_penalty,
+            "repetition_penalty": repetition_penalty,
         }
         if lora is not None:
             body["lora"] = lora
+        if stop is not None:
+            body["stop"] = stop if isinstance(stop, str) else list(stop)
+        if seed is not None:
+            body["seed"] = seed
+        if logprobs is not None:
+            body["logprobs"] = logprobs
         return self._post("/v1/completions", json=body).json()
 
     def completion_stream(
