@@ -32,6 +32,17 @@ def test_client_completion_and_usage(gw):
     assert len(out["choices"][0]["token_ids"]) == 3
 
 
+def test_client_sampling_surface(gw):
+    """seed reproduces, logprobs shape present, penalties accepted."""
+    a = gw.completion("m", "abc", max_tokens=4, temperature=0.7, seed=9,
+                      logprobs=1, presence_penalty=0.1)
+    b = gw.completion("m", "abc", max_tokens=4, temperature=0.7, seed=9,
+                      logprobs=1, presence_penalty=0.1)
+    assert a["choices"][0]["token_ids"] == b["choices"][0]["token_ids"]
+    lp = a["choices"][0]["logprobs"]
+    assert len(lp["token_logprobs"]) == 4 and len(lp["top_logprobs"]) == 4
+
+
 def test_client_admin_roundtrip(gw):
     gw.register_model("extra", criticality="Sheddable")
     names = [m["model_name"] for m in gw.list_inference_models()]
