@@ -379,6 +379,32 @@ class ConfigReconciler:
             except (ConfigError, RuntimeError) as e:
                 log.error("pool membership reconcile failed: %s", e)
             reconcile_adapters(cfg, self.pool)
+            # scheduler thresholds: rebuild the tree when a live edit changes
+            # them (the reference's compile-time consts, made runtime here)
+            last = self.last_config
+            if last is None or (
+                last.kv_cache_threshold,
+                last.queue_threshold_critical,
+                last.queueing_threshold_lora,
+            ) != (
+                cfg.kv_cache_threshold,
+                cfg.queue_threshold_critical,
+                cfg.queueing_threshold_lora,
+            ):
+                sched = getattr(self.pool, "scheduler", None)
+                if sched is not None and hasattr(sched, "set_thresholds"):
+                    sched.set_thresholds(
+                        cfg.kv_cache_threshold,
+                        cfg.queue_threshold_critical,
+                        cfg.queueing_threshold_lora,
+                    )
+                    if last is not None:
+                        log.info(
+                            "scheduler thresholds updated: kv=%.2f qc=%d ql=%d",
+                            cfg.kv_cache_threshold,
+                            cfg.queue_threshold_critical,
+                            cfg.queueing_threshold_lora,
+                        )
         for act in self.remote_actuators:
             reconcile_adapters_remote(cfg, act)
         self.last_config = cfg

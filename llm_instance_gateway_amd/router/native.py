@@ -128,6 +128,17 @@ class NativeScheduler:
         self._state_names = seen
 
     # -- the Scheduler interface ----------------------------------------------
+    def set_thresholds(
+        self,
+        kv_cache_threshold: float,
+        queue_threshold_critical: int,
+        queueing_threshold_lora: int,
+    ) -> None:
+        """Runtime threshold update (mirrors Scheduler.set_thresholds)."""
+        self._r.set_thresholds(
+            kv_cache_threshold, queue_threshold_critical, queueing_threshold_lora
+        )
+
     def schedule(self, req: LLMRequest) -> Engine:
         if self._pool is not None:
             self._sync_pool()

@@ -42,6 +42,23 @@ class Scheduler:
         self.filter = tree if tree is not None else build_default_tree()
         self._rng = rng or random.Random()
 
+    def set_thresholds(
+        self,
+        kv_cache_threshold: float,
+        queue_threshold_critical: int,
+        queueing_threshold_lora: int,
+    ) -> None:
+        """Rebuild the decision tree with new thresholds (runtime config
+        reconcile — the reference carries these as compile-time consts with
+        'make configurable' TODOs, scheduler.go:15-24).  Atomic swap: the
+        tree is immutable once built, so in-flight schedule() calls keep
+        their snapshot."""
+        self.filter = build_default_tree(
+            kv_cache_threshold=kv_cache_threshold,
+            queue_threshold_critical=queue_threshold_critical,
+            queueing_threshold_lora=queueing_threshold_lora,
+        )
+
     def schedule(self, req: LLMRequest) -> Engine:
         """Reference Schedule (scheduler.go:113-122).
 

@@ -272,3 +272,64 @@ def test_reconcile_flips_under_live_traffic(tmp_path, pool):
     assert not errors, errors[:1]
     assert done_count[0] >= 3
     assert {"flip-a", "flip-b"} <= set(h.adapter_registry)
+
+
+def test_runtime_threshold_reconcile(tmp_path, pool):
+    """Live edits to `thresholds:` rebuild the scheduler tree (the
+    reference's compile-time consts, runtime here)."""
+    path = tmp_path / "thr.yaml"
+    path.write_text(
+        "pool: {name: p, base_model: tiny}\n"
+        "thresholds: {queue_threshold_critical: 5}\n"
+    )
+    rec = ConfigReconciler(
+        str(path), pool, poll_interval_s=0.05,
+        health_timeout_s=5, health_interval_s=0.05,
+    )
+    assert rec.reconcile_once()
+    tree_v1 = pool.scheduler.filter
+    # same thresholds -> tree untouched
+    assert rec.reconcile_once()
+    assert pool.scheduler.filter is tree_v1
+    # changed thresholds -> atomic tree swap
+    path.write_text(
+        "pool: {name: p, base_model: tiny}\n"
+        "thresholds: {queue_threshold_critical: 2, kv_cache_threshold: 0.5}\n"
+    )
+    assert rec.reconcile_once()
+    assert pool.scheduler.filter is not tree_v1
+
+
+def test_scheduler_set_thresholds_changes_shed_behavior():
+    """Functional check: tightening queue_threshold_critical flips a
+    sheddable request from routed to dropped."""
+    import pytest as _pytest
+
+    from llm_instance_gateway_amd.router.scheduler import (
+        ResourceExhausted,
+        Scheduler,
+    )
+    from llm_instance_gateway_amd.router.types import (
+        Engine,
+        EngineMetrics,
+        EngineState,
+        LLMRequest,
+    )
+
+    class OneBusyEngine:
+        def all_engine_metrics(self):
+            return [
+                EngineState(
+                    engine=Engine(name="e0", gpu_id=0),
+                    metrics=EngineMetrics(
+                        waiting_queue_size=3, kv_cache_usage_percent=0.5
+                    ),
+                )
+            ]
+
+    sched = Scheduler(OneBusyEngine())
+    shed_req = LLMRequest(model="m", critical=False)
+    assert sched.schedule(shed_req).name == "e0"  # 3 <= default 5
+    sched.set_thresholds(0.8, 2, 50)  # queue gate now 2 < waiting 3
+    with _pytest.raises(ResourceExhausted):
+        sched.schedule(shed_req)
