@@ -162,7 +162,16 @@ def main(argv: Optional[List[str]] = None) -> int:
 
     reconciler = None
     if args.config:
-        reconciler = ConfigReconciler(args.config, pool)
+        reconciler = ConfigReconciler(
+            args.config,
+            pool,
+            health_timeout_s=(
+                cfg.health_check_timeout_s if cfg else 300.0
+            ),
+            health_interval_s=(
+                cfg.health_check_interval_s if cfg else 15.0
+            ),
+        )
         reconciler.start()
 
     app = build_app(pool)
