@@ -27,6 +27,15 @@ from typing import List, Optional
 log = logging.getLogger(__name__)
 
 
+def resolve_port(args_port: int, cfg) -> int:
+    """--port wins when given; else the config's `port:`; else 8081."""
+    if args_port:
+        return args_port
+    if cfg and cfg.port:
+        return cfg.port
+    return 8081
+
+
 def build_pool(args, cfg) -> "EnginePool":
     import torch
 
@@ -78,7 +87,11 @@ def build_pool(args, cfg) -> "EnginePool":
 def main(argv: Optional[List[str]] = None) -> int:
     ap = argparse.ArgumentParser(description=__doc__.splitlines()[0])
     ap.add_argument("--config", default="", help="gateway YAML (pool/models/adapters/thresholds)")
-    ap.add_argument("--port", type=int, default=8081, help="HTTP front-door port (reference gateway.yaml:13-15)")
+    ap.add_argument(
+        "--port", type=int, default=0,
+        help="HTTP front-door port (0 = the config's `port:`, else 8081; "
+        "reference gateway.yaml:13-15)",
+    )
     ap.add_argument("--model", default="", help="base model preset override")
     ap.add_argument("--devices", default="", help="comma list, e.g. cuda:0,cuda:1 or cpu")
     ap.add_argument(
@@ -136,7 +149,7 @@ def main(argv: Optional[List[str]] = None) -> int:
         )
         import uvicorn
 
-        port = args.port if args.port else (cfg.port if cfg else 8081)
+        port = resolve_port(args.port, cfg)
         log.info("remote-router mode over %d engines", len(addrs))
         uvicorn.run(app, host="0.0.0.0", port=port, log_level="warning")
         return 0
@@ -188,7 +201,7 @@ def main(argv: Optional[List[str]] = None) -> int:
 
     import uvicorn
 
-    port = args.port if args.port else (cfg.port if cfg else 8081)
+    port = resolve_port(args.port, cfg)
     uvicorn.run(app, host="0.0.0.0", port=port, log_level="warning")
     pool.stop()
     return 0

@@ -56,3 +56,16 @@ def test_build_pool_defaults_without_config():
         assert pool.datastore.fetch_model("tiny") is not None
     finally:
         pool.stop()
+
+
+def test_port_precedence():
+    """--port wins; else the config's `port:`; else 8081 (a truthy argparse
+    default used to shadow the config value)."""
+    from llm_instance_gateway_amd.api.config import GatewayConfig
+    from llm_instance_gateway_amd.server import resolve_port
+
+    cfg = GatewayConfig()
+    cfg.port = 9099
+    assert resolve_port(0, cfg) == 9099
+    assert resolve_port(7000, cfg) == 7000
+    assert resolve_port(0, None) == 8081
