@@ -193,7 +193,7 @@ def main(argv: Optional[List[str]] = None) -> int:
         log.info("signal %s: draining and stopping", signum)
         if reconciler:
             reconciler.stop()
-        pool.stop()
+        pool.stop(drain_timeout_s=30.0)  # in-flight requests finish
         sys.exit(0)
 
     signal.signal(signal.SIGTERM, shutdown)

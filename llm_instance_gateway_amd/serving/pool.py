@@ -679,8 +679,24 @@ class EnginePool:
             )
             self._watchdog.start()
 
-    def stop(self) -> None:
+    def stop(self, drain_timeout_s: float = 0.0) -> None:
+        """Stop all engines.  ``drain_timeout_s`` > 0: graceful — engines
+        stop ADMITTING (drain on), in-flight requests run to completion (or
+        the deadline), then workers exit.  The reference's SIGTERM grace
+        period (main.go graceful gRPC shutdown)."""
         self._stop.set()
+        if drain_timeout_s > 0:
+            for h in list(self.handles.values()):
+                try:
+                    h.drain(True)
+                except (BrokenPipeError, OSError):  # already dead
+                    pass
+            deadline = time.monotonic() + drain_timeout_s
+            while time.monotonic() < deadline:
+                with self._tickets_lock:
+                    if not self._tickets:
+                        break
+                time.sleep(0.05)
         for h in list(self.handles.values()):
             h.stop()
 

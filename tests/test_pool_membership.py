@@ -192,3 +192,39 @@ def test_membership_capacity_bound():
     assert base in pool._free_slots
     h2 = pool._make_handle(_spec(2))
     assert h2.slot_base == base
+
+
+def test_graceful_stop_drains_inflight():
+    """pool.stop(drain_timeout_s>0): engines stop admitting but in-flight
+    requests run to completion before workers exit (SIGTERM grace)."""
+    from llm_instance_gateway_amd.router.types import LLMRequest
+    from llm_instance_gateway_amd.serving.pool import EnginePool, PoolRequest
+
+    pool = EnginePool(
+        "tiny",
+        devices=["cpu"],
+        engine_kwargs=dict(
+            num_kv_blocks=64, max_num_seqs=4,
+            max_num_batched_tokens=128, max_model_len=128, max_loras=0,
+        ),
+    )
+    pool.start(watchdog=False)
+    try:
+        tickets = [
+            pool.route_and_submit(
+                LLMRequest(model="tiny", prompt_len=8, max_new_tokens=24),
+                PoolRequest(
+                    request_id=f"drain-{i}",
+                    prompt_token_ids=list(range(4, 12)),
+                    max_new_tokens=24,
+                ),
+            )
+            for i in range(3)
+        ]
+    except Exception:
+        pool.stop()
+        raise
+    pool.stop(drain_timeout_s=60.0)
+    for t in tickets:
+        fin = t.wait(timeout=5)  # already done: drained before workers exited
+        assert len(fin.output_token_ids) == 24
