@@ -82,6 +82,8 @@ class GatewayConfig:
     # provider refresh intervals (reference main.go:38-39)
     refresh_metrics_interval_s: float = 0.050
     refresh_engines_interval_s: float = 10.0
+    # ops: watchdog re-spawns dead engines (kubelet restart semantics)
+    auto_restart_engines: bool = False
     # front door
     port: int = 8081
     health_check_timeout_s: float = DEFAULT_HEALTH_TIMEOUT_S
@@ -138,6 +140,7 @@ def parse_config(raw: Dict[str, Any]) -> GatewayConfig:
     else:
         cfg.ensure_not_exist = [AdapterSpec.parse(a) for a in ene or []]
 
+    cfg.auto_restart_engines = bool(raw.get("auto_restart_engines", False))
     th = raw.get("thresholds", {}) or {}
     cfg.kv_cache_threshold = float(th.get("kv_cache_threshold", cfg.kv_cache_threshold))
     cfg.queue_threshold_critical = int(

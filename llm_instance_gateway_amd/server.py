@@ -155,7 +155,7 @@ def main(argv: Optional[List[str]] = None) -> int:
         return 0
     pool = build_pool(args, cfg)
     log.info("starting %d engine(s)...", len(pool.handles))
-    pool.start()
+    pool.start(auto_restart=bool(cfg and cfg.auto_restart_engines))
 
     if args.native_router:
         from .router.native import NativeScheduler, native_available

@@ -662,7 +662,14 @@ class EnginePool:
         return h
 
     # -- lifecycle ----------------------------------------------------------
-    def start(self, watchdog: bool = True) -> None:
+    def start(self, watchdog: bool = True, auto_restart: bool = False) -> None:
+        """``auto_restart``: the watchdog re-spawns a dead engine (with
+        adapter-registry replay) instead of only dropping it — the kubelet
+        restart semantics the reference's pod fleet gets for free.  Capped
+        at 3 consecutive failures per engine (crash-loop guard); a
+        successful heartbeat resets the count."""
+        self._auto_restart = auto_restart
+        self._restart_failures: Dict[str, int] = {}
         threads = []
         for h in self.handles.values():
             t = threading.Thread(target=h.start, daemon=True)
@@ -708,14 +715,32 @@ class EnginePool:
             for h in list(self.handles.values()):  # membership may change live
                 stale = h.heartbeat_age_s() > HEARTBEAT_STALE_S
                 dead = bool(h.procs) and any(not p.is_alive() for p in h.procs)
+                name = h.engine.name
+                if h.healthy and not (stale or dead):
+                    self._restart_failures.pop(name, None)  # recovered
                 if (stale or dead) and h.healthy:
                     log.error(
                         "engine %s unhealthy (stale=%s dead=%s); dropping from pool",
-                        h.engine.name, stale, dead,
+                        name, stale, dead,
                     )
                     h.healthy = False
-                    self.datastore.remove_engine(h.engine.name)
-                    self._fail_inflight(h.engine.name, "engine died")
+                    self.datastore.remove_engine(name)
+                    self._fail_inflight(name, "engine died")
+                    if getattr(self, "_auto_restart", False):
+                        fails = self._restart_failures.get(name, 0)
+                        if fails >= 3:
+                            log.error(
+                                "engine %s crash-looping (%d restarts); "
+                                "leaving down", name, fails,
+                            )
+                            continue
+                        self._restart_failures[name] = fails + 1
+                        log.warning("auto-restarting engine %s (attempt %d)",
+                                    name, fails + 1)
+                        try:
+                            self.restart_engine(name)
+                        except Exception as e:
+                            log.error("auto-restart of %s failed: %s", name, e)
 
     def _fail_inflight(self, engine_name: str, reason: str) -> None:
         with self._tickets_lock:

@@ -228,3 +228,43 @@ def test_graceful_stop_drains_inflight():
     for t in tickets:
         fin = t.wait(timeout=5)  # already done: drained before workers exited
         assert len(fin.output_token_ids) == 24
+
+
+def test_watchdog_auto_restart_replays_adapters():
+    """auto_restart=True: the watchdog re-spawns a dead engine (kubelet
+    semantics) and the adapter registry is replayed."""
+    import os
+    import time as _time
+
+    from llm_instance_gateway_amd.serving.pool import EnginePool
+
+    pool = EnginePool(
+        "tiny",
+        devices=["cpu"],
+        engine_kwargs=dict(
+            num_kv_blocks=64, max_num_seqs=4,
+            max_num_batched_tokens=128, max_model_len=128, max_loras=2,
+        ),
+        lora_rank=8,
+    )
+    pool.start(watchdog=True, auto_restart=True)
+    try:
+        h = pool.handles["engine-0"]
+        assert h.load_adapter("keep-me", "", 8) is None
+        # murder the worker process out-of-band
+        h.procs[0].terminate()
+        deadline = _time.monotonic() + 30
+        while _time.monotonic() < deadline:
+            if h.healthy and h.alive() and "keep-me" in h.active_adapters:
+                break
+            _time.sleep(0.2)
+        assert h.healthy and h.alive(), "watchdog did not auto-restart"
+        assert "keep-me" in h.adapter_registry
+        deadline = _time.monotonic() + 15
+        while _time.monotonic() < deadline:
+            if "keep-me" in h.active_adapters:
+                break
+            _time.sleep(0.2)
+        assert "keep-me" in h.active_adapters  # replayed onto the new worker
+    finally:
+        pool.stop()
