@@ -552,6 +552,7 @@ def build_app(pool: EnginePool, request_timeout_s: float = 600.0) -> FastAPI:
                 "alive": h.alive(),
                 "draining": h.is_draining(),
                 "heartbeat_age_s": round(h.heartbeat_age_s(), 3),
+                "restarts": pool.restart_counts.get(h.engine.name, 0),
                 "metrics": h.metrics().__dict__,
             }
             for h in pool.handles.values()

@@ -627,6 +627,8 @@ class EnginePool:
         self.scheduler = scheduler or Scheduler(self)
         self._tickets: Dict[str, RequestTicket] = {}
         self._tickets_lock = threading.Lock()
+        # lifetime restarts per engine (observability: /admin/engines)
+        self.restart_counts: Dict[str, int] = {}
         self.on_finished: Optional[Callable[[FinishedRequest], None]] = None
         self._watchdog: Optional[threading.Thread] = None
         self._stop = threading.Event()
@@ -756,6 +758,7 @@ class EnginePool:
     def restart_engine(self, name: str) -> None:
         """Re-spawn a dead engine and replay its adapter registry (reference:
         pod restart + sidecar reconcile, SURVEY.md §5 failure detection)."""
+        self.restart_counts[name] = self.restart_counts.get(name, 0) + 1
         h = self.handles[name]
         h.kill()
         registry = dict(h.adapter_registry)
