@@ -12,9 +12,12 @@ build:
 
 # Debug extension build: host-side index/range validation in the bindings
 # (LIG_DEBUG) — catches bad slot/unit/block-table values before a kernel
-# dereferences them.
+# dereferences them.  Separate --build-temp: setuptools only tracks source
+# mtimes, so sharing objects with `build` would silently reuse the other
+# configuration's .o files.  Run `make build` afterwards to restore the
+# release .so in-tree (the snapshot ships whatever is in-tree).
 build-debug:
-	LIG_DEBUG=1 PYTORCH_ROCM_ARCH=$(ARCH) $(PYTHON) setup.py build_ext --inplace
+	LIG_DEBUG=1 PYTORCH_ROCM_ARCH=$(ARCH) $(PYTHON) setup.py build_ext --inplace --build-temp build/debug-temp
 
 # Sanitizer pass over the native router core (production logic under
 # ASan+UBSan; SURVEY.md §5 names this gap in the reference).
