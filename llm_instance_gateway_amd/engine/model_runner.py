@@ -112,7 +112,18 @@ class ModelRunner:
         self.graph_runner = None
         # MoE decode is graph-safe too: decode-only batches use the dense
         # expert dispatch (static shapes, no host sync — models/mixtral.py).
-        self._use_graphs = self.device.type == "cuda" and not config.enforce_eager
+        # TP engines decode EAGER by default: capturing the per-block RCCL
+        # all-reduce inside a hipGraph is exactly the stream-semantics
+        # surface that has never run on multi-GPU hardware (docs/scaling.md)
+        # — conservative by construction for the first TP run; LIG_TP_GRAPHS=1
+        # opts in once validated.
+        import os as _os
+
+        self._use_graphs = (
+            self.device.type == "cuda"
+            and not config.enforce_eager
+            and (tp is None or _os.environ.get("LIG_TP_GRAPHS", "0") == "1")
+        )
         # Mixed-step prefill/decode stream overlap — MEASURED DEAD END as a
         # default: 51.4 vs 56.5 req/s at the serving shape (gpurun
         # bench_mixed vs bench_mixed_noov) — the graph replay and the eager
@@ -120,8 +131,6 @@ class ModelRunner:
         # streams only added sync overhead; the single flat mixed batch
         # already keeps the chip full.  Kept opt-in (LIG_OVERLAP=1) for
         # shapes where decode is small vs prefill.
-        import os as _os
-
         self._overlap = _os.environ.get("LIG_OVERLAP", "0") == "1"
         self._decode_stream = None
 
