@@ -460,11 +460,20 @@ class EngineHandle:
         self._reader.start()
 
     def _read_loop(self) -> None:
+        # Bind the connection ONCE: this thread serves exactly the pipe it
+        # was started for.  Re-reading self.conn each iteration raced with
+        # restart_engine — the superseded reader could consume messages from
+        # the NEW worker's pipe, and its late EOFError flipped healthy=False
+        # on the freshly restarted engine (rare flake caught by soak runs).
+        conn = self.conn
         while True:
             try:
-                msg = self.conn.recv()
+                msg = conn.recv()
             except (EOFError, OSError):
-                self.healthy = False
+                if self.conn is conn:
+                    # the live connection died -> the engine really is gone
+                    self.healthy = False
+                # else: a restart superseded this reader; exit silently
                 return
             op = msg[0]
             if op == "finished":
