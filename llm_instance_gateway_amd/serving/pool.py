@@ -729,6 +729,17 @@ class EnginePool:
                 name = h.engine.name
                 if h.healthy and not (stale or dead):
                     self._restart_failures.pop(name, None)  # recovered
+                if not h.healthy and not stale and not dead and h.procs:
+                    # Heartbeats resumed and the workers are alive: the
+                    # engine was only STALLED (host CPU starvation, long
+                    # kernel), not dead — return it to the routing set (the
+                    # reference's EndpointSlice Ready condition flipping
+                    # back to Ready).  Crash recovery still goes through
+                    # restart_engine (procs dead ⇒ `dead` stays true).
+                    log.warning("engine %s heartbeat recovered; re-adding", name)
+                    h.healthy = True
+                    self.datastore.add_engine(h.engine)
+                    continue
                 if (stale or dead) and h.healthy:
                     log.error(
                         "engine %s unhealthy (stale=%s dead=%s); dropping from pool",

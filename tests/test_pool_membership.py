@@ -270,3 +270,48 @@ def test_watchdog_auto_restart_replays_adapters():
         assert "keep-me" in h.active_adapters  # replayed onto the new worker
     finally:
         pool.stop()
+
+
+def test_watchdog_recovers_stalled_engine(monkeypatch):
+    """A STALLED engine (heartbeats paused, workers alive — host CPU
+    starvation / long kernel) is dropped from the routing set and then
+    RE-ADDED when heartbeats resume (EndpointSlice Ready flipping back),
+    without a restart."""
+    import os
+    import signal
+    import time as _time
+
+    import llm_instance_gateway_amd.serving.pool as pool_mod
+    from llm_instance_gateway_amd.serving.pool import EnginePool
+
+    monkeypatch.setattr(pool_mod, "HEARTBEAT_STALE_S", 1.0)
+    pool = EnginePool(
+        "tiny",
+        devices=["cpu"],
+        engine_kwargs=dict(
+            num_kv_blocks=64, max_num_seqs=4,
+            max_num_batched_tokens=128, max_model_len=128, max_loras=0,
+        ),
+    )
+    pool.start(watchdog=True)
+    try:
+        h = pool.handles["engine-0"]
+        pid = h.procs[0].pid
+        os.kill(pid, signal.SIGSTOP)  # freeze: heartbeats stop, proc alive
+        try:
+            deadline = _time.monotonic() + 30
+            while _time.monotonic() < deadline and h.healthy:
+                _time.sleep(0.1)
+            assert not h.healthy, "stalled engine was not dropped"
+            assert pool.datastore.get_engine("engine-0") is None
+        finally:
+            os.kill(pid, signal.SIGCONT)
+        deadline = _time.monotonic() + 30
+        while _time.monotonic() < deadline and not h.healthy:
+            _time.sleep(0.1)
+        assert h.healthy, "recovered engine was not re-added"
+        assert pool.datastore.get_engine("engine-0") is not None
+        assert h.procs[0].is_alive()  # same worker, no restart
+        assert pool.restart_counts.get("engine-0", 0) == 0
+    finally:
+        pool.stop()
