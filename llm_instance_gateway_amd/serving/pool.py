@@ -401,6 +401,9 @@ class EngineHandle:
         # interleave and corrupt (kills the worker under adapter rollout +
         # live traffic).
         self._send_lock = threading.Lock()
+        # True once the pool has decided this engine should STOP (scale-in /
+        # shutdown): the watchdog must never resurrect a retired engine.
+        self.retired = False
 
     def _send(self, msg) -> None:
         with self._send_lock:
@@ -453,6 +456,7 @@ class EngineHandle:
             raise TimeoutError(f"engine {self.engine.name} did not become ready")
         msg = parent_conn.recv()
         assert msg[0] == "ready", msg
+        self.retired = False  # (re)started deliberately
         self.healthy = True
         self._reader = threading.Thread(
             target=self._read_loop, daemon=True, name=f"reader-{self.engine.name}"
@@ -536,6 +540,7 @@ class EngineHandle:
         self._send(("drain", on))
 
     def stop(self) -> None:
+        self.retired = True
         if self.conn is not None:
             try:
                 self._send(("stop",))
@@ -729,7 +734,10 @@ class EnginePool:
                 name = h.engine.name
                 if h.healthy and not (stale or dead):
                     self._restart_failures.pop(name, None)  # recovered
-                if not h.healthy and not stale and not dead and h.procs:
+                if (
+                    not h.healthy and not stale and not dead and h.procs
+                    and not h.retired
+                ):
                     # Heartbeats resumed and the workers are alive: the
                     # engine was only STALLED (host CPU starvation, long
                     # kernel), not dead — return it to the routing set (the
@@ -811,6 +819,7 @@ class EnginePool:
         h = self.handles.get(name)
         if h is None:
             return
+        h.retired = True  # watchdog must not re-add during drain/stop
         self.datastore.remove_engine(name)
         if drain and h.alive():
             h.drain(True)
