@@ -756,7 +756,17 @@ class EnginePool:
                     h.healthy = False
                     self.datastore.remove_engine(name)
                     self._fail_inflight(name, "engine died")
-                    if getattr(self, "_auto_restart", False):
+                # Auto-restart is gated on the OBSERVED DOWN STATE, not on
+                # the healthy->unhealthy transition above: the pipe reader
+                # thread usually notices a dead worker first (EOF within ms)
+                # and flips healthy itself, so a transition-gated restart
+                # would ~never fire.
+                if (
+                    getattr(self, "_auto_restart", False)
+                    and not h.healthy
+                    and not h.retired
+                    and (stale or dead)
+                ):
                         fails = self._restart_failures.get(name, 0)
                         if fails >= 3:
                             log.error(

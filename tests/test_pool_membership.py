@@ -251,21 +251,30 @@ def test_watchdog_auto_restart_replays_adapters():
     try:
         h = pool.handles["engine-0"]
         assert h.load_adapter("keep-me", "", 8) is None
-        # murder the worker process out-of-band
-        h.procs[0].terminate()
+        # murder the worker process out-of-band and WAIT for the death to
+        # be observed (multiprocessing is_alive() lags the signal, so a
+        # naive healthy-check here passes vacuously against the old proc)
+        victim = h.procs[0]
+        victim.terminate()
+        victim.join(timeout=60)
+        assert not victim.is_alive()
         # generous deadlines: a fresh spawn re-imports torch, which can take
         # minutes on a heavily loaded machine (e.g. 4-way pytest-xdist with
         # several engine pools spawning concurrently)
         deadline = _time.monotonic() + 300
         while _time.monotonic() < deadline:
-            if h.healthy and h.alive() and "keep-me" in h.active_adapters:
+            if (
+                pool.restart_counts.get("engine-0", 0) >= 1
+                and h.healthy
+                and h.alive()
+            ):
                 break
             _time.sleep(0.2)
+        assert pool.restart_counts.get("engine-0", 0) >= 1, "no restart fired"
         assert h.healthy and h.alive(), (
-            "watchdog did not auto-restart: "
-            f"restarts={pool.restart_counts.get('engine-0', 0)} "
-            f"healthy={h.healthy} alive={h.alive()}"
+            f"restart incomplete: healthy={h.healthy} alive={h.alive()}"
         )
+        assert h.procs[0] is not victim  # genuinely re-spawned
         assert "keep-me" in h.adapter_registry
         deadline = _time.monotonic() + 120
         while _time.monotonic() < deadline:
